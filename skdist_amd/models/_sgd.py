@@ -1,0 +1,431 @@
+"""
+Batched mini-batch-SGD linear solver — the engine behind
+``skdist_amd.models.LogisticRegression`` / ``LinearSVC`` / ``Ridge``.
+
+Design (MI355X-first):
+
+  * ONE solver instance trains ``ncols`` independent linear models at once:
+    every (candidate × CV-fold × class) combination becomes one column of a
+    weight matrix ``W [f+1, ncols]`` (last row = intercept; X carries an
+    implicit ones-column).  A 500-candidate × 5-fold × binary search is a
+    2500-column batch — two MFMA GEMMs per mini-batch step instead of 2500
+    separate tiny fits.
+
+  * per-column metadata drives the fused epilogues:
+      - ``col_fold``  : the CV fold this column must NOT train on
+        (samples with ``fold_id[i] == col_fold[c]`` get gradient 0);
+      - ``col_class`` : target class for internal one-vs-rest multiclass
+        (binary targets are class 1 of 2);
+      - ``col_lr`` / ``col_l2``: per-column hyper-parameters.
+
+  * the same code runs three ways:
+      - torch CPU fp32 — the ``sc=None`` path and the numerics reference;
+      - torch GPU — eager fallback (tests only; the GPU path refuses to
+        run eagerly unless SKDIST_AMD_ALLOW_EAGER=1);
+      - hand-written HIP kernels (skdist_amd.ops) — the production path:
+        fused forward GEMM+σ+mask and backward GEMMᵀ+update on MFMA.
+
+Numerics: X is held bf16 on GPU (fp32 on CPU), W master weights fp32,
+GEMM accumulation fp32 (MFMA native).  Gradient for column c:
+
+    g[i,c]   = mask[i,c] * dloss(z[i,c], target[i,c])          # forward
+    grad[:,c] = Xᵀ g[:,c] / m_batch + l2[c] * W[:,c]           # backward
+    W[:,c]  -= lr[c] * grad[:,c]                                # update
+
+dloss: logistic  σ(z) - t
+       hinge     -t' * 1[t'·z < 1]      (t' = 2t-1 ∈ {-1, +1})
+       squared   z - t
+"""
+
+import os
+
+import numpy as np
+import torch
+
+LOSS_LOG, LOSS_HINGE, LOSS_SQUARED = 0, 1, 2
+_LOSS_IDS = {"log": LOSS_LOG, "hinge": LOSS_HINGE, "squared": LOSS_SQUARED}
+
+
+def _use_hip(device):
+    """HIP kernels are mandatory on GPU unless explicitly waived."""
+    if device.type != "cuda":
+        return False
+    if os.environ.get("SKDIST_AMD_ALLOW_EAGER") == "1":
+        from ..ops import hip_available
+
+        return hip_available()
+    from ..ops import require_hip
+
+    require_hip()  # raises if the extension is missing on a GPU box
+    return True
+
+
+class DeviceDataset:
+    """(X, y) resident on one device, shared by every fit in a search.
+
+    Holds ``Xaug`` = [X_standardized | 1] as one contiguous matrix
+    (bf16 on GPU, fp32 on CPU), the integer-encoded labels, and the
+    per-sample CV fold id.  On a distributed Cluster the tensors are
+    broadcast ONCE from rank 0 over RCCL/xGMI (reference analog: Spark
+    broadcast at search.py:411, minus the 2 GB workaround).
+    """
+
+    def __init__(self, X, y, cluster=None, device=None, standardize=True,
+                 classes=None):
+        self.cluster = cluster
+        if device is None:
+            device = cluster.device if cluster is not None else (
+                torch.device("cuda") if torch.cuda.is_available()
+                else torch.device("cpu")
+            )
+        self.device = torch.device(device)
+        comp_dtype = (
+            torch.bfloat16 if self.device.type == "cuda" else torch.float32
+        )
+        self.comp_dtype = comp_dtype
+
+        if cluster is not None and cluster.distributed:
+            Xt = None
+            yt = None
+            if X is not None:
+                Xt = torch.as_tensor(np.ascontiguousarray(X), dtype=torch.float32)
+            if y is not None:
+                yt = torch.as_tensor(np.ascontiguousarray(y))
+            Xt = cluster.bcast_tensor(
+                Xt.to(cluster.device) if Xt is not None else None
+            )
+            yt = cluster.bcast_tensor(
+                yt.to(cluster.device) if yt is not None else None
+            )
+        else:
+            Xt = torch.as_tensor(
+                np.ascontiguousarray(X), dtype=torch.float32
+            ).to(self.device)
+            yt = torch.as_tensor(np.ascontiguousarray(y)).to(self.device)
+
+        self.n, self.f = Xt.shape
+        # label encoding (host-visible classes, device int targets)
+        if yt.dtype.is_floating_point and classes is None:
+            # regression targets
+            self.classes_ = None
+            self.y_float = yt.to(torch.float32)
+            self.y_int = None
+        else:
+            y_np = yt.cpu().numpy()
+            self.classes_ = (
+                np.asarray(classes) if classes is not None else np.unique(y_np)
+            )
+            lut = {v: i for i, v in enumerate(self.classes_)}
+            enc = np.asarray([lut[v] for v in y_np], dtype=np.int32)
+            self.y_int = torch.as_tensor(enc, device=self.device)
+            self.y_float = self.y_int.to(torch.float32)
+
+        # standardize + augment with ones column (intercept)
+        if standardize:
+            mean = Xt.mean(dim=0)
+            std = Xt.std(dim=0, unbiased=False)
+            std = torch.where(std > 1e-12, std, torch.ones_like(std))
+            Xt = (Xt - mean) / std
+            self.feat_mean = mean.cpu().numpy()
+            self.feat_std = std.cpu().numpy()
+        else:
+            self.feat_mean = np.zeros(self.f, dtype=np.float32)
+            self.feat_std = np.ones(self.f, dtype=np.float32)
+        ones = torch.ones(self.n, 1, dtype=Xt.dtype, device=self.device)
+        self.Xaug = torch.cat([Xt, ones], dim=1).to(comp_dtype).contiguous()
+        del Xt
+
+        self.fold_id = None  # set by set_cv_partition
+
+    def set_cv_partition(self, cv_splits):
+        """Encode partition-style CV splits as a per-sample fold id.
+
+        Returns False when the splits do not partition the sample set
+        (e.g. ShuffleSplit) — callers then fall back to the generic path.
+        """
+        if not cv_splits:
+            self.fold_id = torch.full(
+                (self.n,), -1, dtype=torch.int32, device=self.device
+            )
+            return True
+        fold = np.full(self.n, -1, dtype=np.int32)
+        for k, (_, test_idx) in enumerate(cv_splits):
+            if np.any(fold[test_idx] != -1):
+                return False
+            fold[test_idx] = k
+        if np.any(fold == -1):
+            return False
+        self.fold_id = torch.as_tensor(fold, device=self.device)
+        return True
+
+    def unstandardize_coef(self, w, b):
+        """Map standardized-space (w, b) back to raw-feature space."""
+        w_raw = w / self.feat_std
+        b_raw = b - np.dot(w_raw, self.feat_mean)
+        return w_raw, b_raw
+
+
+class ColumnSpec:
+    """Per-column metadata for one batched solve (device tensors)."""
+
+    def __init__(self, device, col_fold, col_class, col_lr, col_l2):
+        as_t = lambda a, dt: torch.as_tensor(
+            np.ascontiguousarray(a), dtype=dt, device=device
+        )
+        self.col_fold = as_t(col_fold, torch.int32)
+        self.col_class = as_t(col_class, torch.int32)
+        self.col_lr = as_t(col_lr, torch.float32)
+        self.col_l2 = as_t(col_l2, torch.float32)
+        self.ncols = len(col_fold)
+
+
+def batched_sgd_fit(ds, spec, loss, epochs, batch_size, seed=0,
+                    momentum=0.0, lr_decay=0.0, force_eager=False):
+    """Train all columns; returns W [f+1, ncols] fp32 on ds.device.
+
+    Mini-batches walk a fixed seeded permutation of the rows (re-drawn
+    per epoch on the host RNG, applied as a device gather).
+    """
+    device = ds.device
+    n, fa = ds.Xaug.shape
+    W = torch.zeros(fa, spec.ncols, dtype=torch.float32, device=device)
+    V = (
+        torch.zeros_like(W) if momentum > 0.0 else None
+    )
+    hip = (not force_eager) and _use_hip(device)
+
+    rng = np.random.default_rng(seed)
+    loss_id = _LOSS_IDS[loss] if isinstance(loss, str) else loss
+
+    target_base = ds.y_float if ds.y_int is None else None
+
+    for epoch in range(epochs):
+        perm = torch.as_tensor(
+            rng.permutation(n), dtype=torch.int64, device=device
+        )
+        lr_scale = 1.0 / (1.0 + lr_decay * epoch)
+        for start in range(0, n, batch_size):
+            idx = perm[start : start + batch_size]
+            if hip:
+                from ..ops import sgd_step_hip
+
+                sgd_step_hip(
+                    ds.Xaug, ds.y_float, ds.fold_id, idx, W, V, spec,
+                    loss_id, lr_scale, momentum,
+                )
+            else:
+                _sgd_step_torch(
+                    ds.Xaug, ds.y_float, ds.fold_id, idx, W, V, spec,
+                    loss_id, lr_scale, momentum,
+                )
+    return W
+
+
+def _sgd_step_torch(Xaug, y_float, fold_id, idx, W, V, spec, loss_id,
+                    lr_scale, momentum):
+    """One mini-batch step, eager torch — the numerics reference the HIP
+    kernels are tested against (fp32 on CPU; on GPU it mirrors the kernel's
+    bf16-in/fp32-accumulate)."""
+    Xb = Xaug[idx]                       # [m, f+1]
+    m = Xb.shape[0]
+    comp = Xb.dtype
+    Z = (Xb @ W.to(comp)).to(torch.float32)      # [m, ncols] fp32 accum
+
+    # targets: (y == col_class) as {0,1}; regression uses y directly
+    yb = y_float[idx]
+    t = (yb.unsqueeze(1) == spec.col_class.unsqueeze(0).to(torch.float32))
+    t = t.to(torch.float32)
+    if loss_id == LOSS_SQUARED and spec.col_class[0].item() < 0:
+        t = yb.unsqueeze(1).expand_as(Z)
+
+    if loss_id == LOSS_LOG:
+        G = torch.sigmoid(Z) - t
+    elif loss_id == LOSS_HINGE:
+        s = 2.0 * t - 1.0
+        G = torch.where(s * Z < 1.0, -s, torch.zeros_like(Z))
+    else:
+        G = Z - t
+
+    if fold_id is not None:
+        mask = (
+            fold_id[idx].unsqueeze(1) != spec.col_fold.unsqueeze(0)
+        ).to(torch.float32)
+        G = G * mask
+
+    grad = (Xb.transpose(0, 1).to(comp) @ G.to(comp)).to(torch.float32) / m
+    # L2 on weights only, not the intercept row
+    grad[:-1] += spec.col_l2.unsqueeze(0) * W[:-1]
+    step = spec.col_lr.unsqueeze(0) * lr_scale * grad
+    if momentum > 0.0:
+        V.mul_(momentum).add_(step)
+        W.sub_(V)
+    else:
+        W.sub_(step)
+
+
+# --------------------------------------------------------------------- #
+# batched scoring
+# --------------------------------------------------------------------- #
+
+def batched_scores(ds, spec, W, loss_id, col_model, n_models, n_classes,
+                   metric, chunk=262144, force_eager=False):
+    """Per-model test-fold metric for every model in the batch.
+
+    ``col_model`` maps each column to its model id; a model owns 1 column
+    (binary / regression) or ``n_classes`` consecutive columns (internal
+    OvR).  The test rows of model m are those with
+    ``fold_id[i] == col_fold[first column of m]``.
+
+    Streams X in row chunks: Z-chunk GEMM → per-model sufficient
+    statistics (correct counts / confusion / log-loss sums / AUC
+    histograms), so the n×ncols score matrix never materializes.
+    Returns np.ndarray [n_models] of metric values.
+    """
+    device = ds.device
+    n = ds.n
+    ncols = spec.ncols
+    first_col = torch.as_tensor(
+        np.arange(n_models, dtype=np.int64) * (n_classes if n_classes > 2 else 1),
+        device=device,
+    )
+    model_fold = spec.col_fold[first_col]          # [n_models]
+
+    stats = _MetricState(metric, n_models, n_classes, device)
+    Wc = W.to(ds.comp_dtype)
+    for start in range(0, n, chunk):
+        end = min(n, start + chunk)
+        Xb = ds.Xaug[start:end]
+        Z = (Xb @ Wc).to(torch.float32)            # [m, ncols]
+        fid = ds.fold_id[start:end]
+        yb = ds.y_float[start:end]
+        stats.update(Z, yb, fid, spec, model_fold, n_classes)
+    return stats.finalize()
+
+
+class _MetricState:
+    """Streaming sufficient statistics for batched test-fold metrics."""
+
+    N_AUC_BINS = 8192
+
+    def __init__(self, metric, n_models, n_classes, device):
+        self.metric = metric
+        self.n_models = n_models
+        self.n_classes = max(n_classes, 2)
+        self.device = device
+        z = lambda *shape: torch.zeros(*shape, dtype=torch.float64,
+                                       device=device)
+        if metric in ("accuracy", "f1", "f1_weighted", "f1_macro"):
+            k = self.n_classes
+            self.confusion = z(n_models, k, k)  # [model, true, pred]
+        elif metric == "neg_log_loss":
+            self.loss_sum = z(n_models)
+            self.count = z(n_models)
+        elif metric == "roc_auc":
+            self.pos_hist = z(n_models, self.N_AUC_BINS)
+            self.neg_hist = z(n_models, self.N_AUC_BINS)
+        elif metric in ("r2", "neg_mean_squared_error"):
+            self.sse = z(n_models)
+            self.sy = z(n_models)
+            self.syy = z(n_models)
+            self.count = z(n_models)
+        else:
+            raise ValueError(f"unsupported device metric: {metric}")
+
+    def update(self, Z, yb, fid, spec, model_fold, n_classes):
+        m = Z.shape[0]
+        dev = self.device
+        nm = self.n_models
+        # test mask per model: fold_id == model_fold  [m, n_models]
+        tmask = fid.unsqueeze(1) == model_fold.unsqueeze(0)
+        if n_classes > 2:
+            Zm = Z.view(m, nm, n_classes)
+            pred = Zm.argmax(dim=2)                        # [m, nm]
+        else:
+            pred = (Z >= 0).to(torch.int64)                # [m, nm]
+
+        if self.metric in ("accuracy", "f1", "f1_weighted", "f1_macro"):
+            k = self.n_classes
+            true = yb.to(torch.int64).unsqueeze(1).expand(m, nm)
+            flat = (
+                torch.arange(nm, device=dev).unsqueeze(0).expand(m, nm) * k * k
+                + true * k + pred
+            )
+            self.confusion.view(-1).scatter_add_(
+                0, flat[tmask].view(-1),
+                torch.ones(int(tmask.sum()), dtype=torch.float64, device=dev),
+            )
+        elif self.metric == "neg_log_loss":
+            if n_classes > 2:
+                Zm = Z.view(m, nm, n_classes)
+                logp = Zm.log_softmax(dim=2)
+                true = yb.to(torch.int64)
+                ll = -logp.gather(
+                    2, true.view(m, 1, 1).expand(m, nm, 1)
+                ).squeeze(2)
+            else:
+                t = yb.unsqueeze(1)
+                ll = torch.nn.functional.softplus(Z) - t * Z
+            ll = torch.where(tmask, ll.to(torch.float64), torch.zeros_like(ll, dtype=torch.float64))
+            self.loss_sum += ll.sum(dim=0)
+            self.count += tmask.sum(dim=0).to(torch.float64)
+        elif self.metric == "roc_auc":
+            p = torch.sigmoid(Z)
+            bins = (p * (self.N_AUC_BINS - 1)).to(torch.int64)
+            t = yb.unsqueeze(1) == spec.col_class.unsqueeze(0).to(torch.float32)
+            base = torch.arange(nm, device=dev).unsqueeze(0) * self.N_AUC_BINS
+            flat = (base + bins)
+            ones = torch.ones(1, dtype=torch.float64, device=dev)
+            posm = tmask & t
+            negm = tmask & ~t
+            self.pos_hist.view(-1).scatter_add_(
+                0, flat[posm].view(-1), ones.expand(int(posm.sum()))
+            )
+            self.neg_hist.view(-1).scatter_add_(
+                0, flat[negm].view(-1), ones.expand(int(negm.sum()))
+            )
+        else:  # regression
+            t = yb.unsqueeze(1)
+            err = (Z - t).to(torch.float64)
+            tm = tmask.to(torch.float64)
+            self.sse += (err * err * tm).sum(dim=0)
+            self.sy += (t.to(torch.float64) * tm).sum(dim=0)
+            self.syy += (t.to(torch.float64) ** 2 * tm).sum(dim=0)
+            self.count += tm.sum(dim=0)
+
+    def finalize(self):
+        if self.metric == "accuracy":
+            conf = self.confusion
+            correct = conf.diagonal(dim1=1, dim2=2).sum(dim=1)
+            total = conf.sum(dim=(1, 2)).clamp_min(1)
+            return (correct / total).cpu().numpy()
+        if self.metric in ("f1", "f1_weighted", "f1_macro"):
+            conf = self.confusion
+            tp = conf.diagonal(dim1=1, dim2=2)              # [nm, k]
+            support = conf.sum(dim=2)                       # true counts
+            pred_ct = conf.sum(dim=1)
+            prec = tp / pred_ct.clamp_min(1e-12)
+            rec = tp / support.clamp_min(1e-12)
+            f1 = 2 * prec * rec / (prec + rec).clamp_min(1e-12)
+            if self.metric == "f1":
+                return f1[:, 1].cpu().numpy()
+            if self.metric == "f1_macro":
+                return f1.mean(dim=1).cpu().numpy()
+            w = support / support.sum(dim=1, keepdim=True).clamp_min(1)
+            return (f1 * w).sum(dim=1).cpu().numpy()
+        if self.metric == "neg_log_loss":
+            return (-(self.loss_sum / self.count.clamp_min(1))).cpu().numpy()
+        if self.metric == "roc_auc":
+            # AUC from score histograms: P(score_pos > score_neg) + 0.5 ties
+            pos = self.pos_hist
+            neg = self.neg_hist
+            neg_cum = neg.cumsum(dim=1) - neg  # negatives strictly below bin
+            auc = (pos * (neg_cum + 0.5 * neg)).sum(dim=1)
+            denom = (pos.sum(dim=1) * neg.sum(dim=1)).clamp_min(1)
+            return (auc / denom).cpu().numpy()
+        if self.metric == "neg_mean_squared_error":
+            return (-(self.sse / self.count.clamp_min(1))).cpu().numpy()
+        if self.metric == "r2":
+            mean = self.sy / self.count.clamp_min(1)
+            sst = self.syy - self.count * mean * mean
+            return (1.0 - self.sse / sst.clamp_min(1e-12)).cpu().numpy()
+        raise AssertionError

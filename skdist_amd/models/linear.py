@@ -1,0 +1,339 @@
+"""
+GPU-native linear estimators with the sklearn API.
+
+These replace the liblinear/lbfgs solvers sk-dist leaned on (reference
+SURVEY.md §2.4 row 1).  ``fit`` trains one model; ``batched_cv_fit_score``
+is the protocol hook :class:`skdist_amd.distribute.search.DistBaseSearchCV`
+uses to train EVERY (candidate × fold) model in one batched solve on the
+GPU cluster.
+
+Fitted state is host numpy only (``coef_``, ``intercept_``, ``classes_``)
+— models pickle and predict exactly like sklearn estimators (sk-dist
+contract, reference search.py:568-570).
+"""
+
+import time
+
+import numpy as np
+from sklearn.base import BaseEstimator, ClassifierMixin, RegressorMixin
+
+from ._sgd import (
+    LOSS_HINGE,
+    LOSS_LOG,
+    LOSS_SQUARED,
+    ColumnSpec,
+    DeviceDataset,
+    batched_scores,
+    batched_sgd_fit,
+)
+
+
+class FallbackToGeneric(Exception):
+    """Raised when a batched solve cannot honor the request; the search
+    falls back to the per-task generic path."""
+
+
+_DEVICE_METRICS = {
+    "accuracy", "f1", "f1_weighted", "f1_macro", "neg_log_loss", "roc_auc",
+    "r2", "neg_mean_squared_error",
+}
+
+
+class _BatchedLinearBase(BaseEstimator):
+    """Shared machinery for the SGD-trained linear family."""
+
+    _loss = LOSS_LOG  # overridden
+
+    def _lam(self, n_train):
+        """Per-column L2 coefficient from the sklearn-style hyper-param."""
+        C = getattr(self, "C", None)
+        if C is not None:
+            return 1.0 / (float(C) * max(n_train, 1))
+        return float(getattr(self, "alpha", 1.0)) / max(n_train, 1)
+
+    def _hyper_names(self):
+        return {"C"} if hasattr(self, "C") else {"alpha"}
+
+    # ------------------------------------------------------------------ #
+    def fit(self, X, y, sample_weight=None):
+        if sample_weight is not None:
+            raise ValueError("sample_weight is not supported yet")
+        t0 = time.perf_counter()
+        sc = getattr(self, "sc", None)
+        is_clf = isinstance(self, ClassifierMixin)
+        ds = DeviceDataset(
+            np.asarray(X, dtype=np.float32), np.asarray(y),
+            cluster=None,  # single fit: local device, no broadcast
+            device=sc.device if sc is not None else None,
+            standardize=self.standardize,
+        )
+        ds.set_cv_partition([])  # no CV mask
+        if is_clf:
+            k = len(ds.classes_)
+            ncols = 1 if k == 2 else k
+            col_class = (
+                np.array([1], dtype=np.int32) if k == 2
+                else np.arange(k, dtype=np.int32)
+            )
+        else:
+            ncols = 1
+            col_class = np.array([-1], dtype=np.int32)
+        lam = self._lam(ds.n)
+        spec = ColumnSpec(
+            ds.device,
+            col_fold=np.full(ncols, -2, dtype=np.int32),
+            col_class=col_class,
+            col_lr=np.full(ncols, self.lr, dtype=np.float32),
+            col_l2=np.full(ncols, lam, dtype=np.float32),
+        )
+        W = batched_sgd_fit(
+            ds, spec, self._loss, self.epochs, self.batch_size,
+            seed=self._seed(), momentum=self.momentum,
+        )
+        self._store_fitted(ds, W, is_clf)
+        self.n_features_in_ = ds.f
+        self.fit_time_ = time.perf_counter() - t0
+        return self
+
+    def _seed(self):
+        rs = getattr(self, "random_state", None)
+        return 0 if rs is None else int(rs)
+
+    def _store_fitted(self, ds, W, is_clf):
+        Wh = W.cpu().numpy()  # [f+1, ncols]
+        w_std = Wh[:-1].T     # [ncols, f] standardized space
+        b_std = Wh[-1]
+        w_raw = np.empty_like(w_std)
+        b_raw = np.empty_like(b_std)
+        for c in range(w_std.shape[0]):
+            w_raw[c], b_raw[c] = ds.unstandardize_coef(w_std[c], b_std[c])
+        if is_clf:
+            self.classes_ = ds.classes_
+            self.coef_ = w_raw
+            self.intercept_ = b_raw
+        else:
+            self.coef_ = w_raw[0]
+            self.intercept_ = float(b_raw[0])
+
+    # ------------------------------------------------------------------ #
+    # host-side inference (numpy, sklearn-compatible)
+    # ------------------------------------------------------------------ #
+    def decision_function(self, X):
+        X = np.asarray(X, dtype=np.float64)
+        scores = X @ self.coef_.T + self.intercept_
+        if scores.ndim == 2 and scores.shape[1] == 1:
+            return scores.ravel()
+        return scores
+
+    # ------------------------------------------------------------------ #
+    # the batched-search protocol (used by DistBaseSearchCV.fit)
+    # ------------------------------------------------------------------ #
+    def batched_cv_fit_score(self, X, y, candidate_params, cv_splits,
+                             scoring, scorers, cluster,
+                             return_train_score=False):
+        """Train all (candidate × fold) models in one batched device solve.
+
+        Returns the same per-task result dicts the generic path produces
+        (task order: candidates outer, folds inner) so ``_format_results``
+        is shared.  Raises :class:`FallbackToGeneric` when the request
+        can't be batched (unsupported params / scoring / non-partition CV).
+        """
+        if return_train_score:
+            raise FallbackToGeneric("return_train_score not batched yet")
+        metric = self._device_metric(scoring)
+        unsupported = set().union(
+            *(set(p) for p in candidate_params)
+        ) - self._hyper_names() - {"lr"}
+        if unsupported:
+            raise FallbackToGeneric(f"non-batchable params {unsupported}")
+
+        is_clf = isinstance(self, ClassifierMixin)
+        t0 = time.perf_counter()
+        ds = DeviceDataset(
+            None if X is None else np.asarray(X, dtype=np.float32),
+            None if y is None else np.asarray(y),
+            cluster=cluster, standardize=self.standardize,
+        )
+        if not ds.set_cv_partition(cv_splits):
+            raise FallbackToGeneric("cv splits do not partition the data")
+        n_folds = len(cv_splits)
+        n_cand = len(candidate_params)
+        fold_train_n = [len(tr) for tr, _ in cv_splits]
+        fold_test_n = [len(te) for _, te in cv_splits]
+
+        if is_clf:
+            k = len(ds.classes_)
+            n_classes = k
+            cols_per_model = 1 if k == 2 else k
+            cls = (
+                np.array([1], dtype=np.int32) if k == 2
+                else np.arange(k, dtype=np.int32)
+            )
+        else:
+            n_classes = 2
+            cols_per_model = 1
+            cls = np.array([-1], dtype=np.int32)
+
+        # this rank's shard of candidates
+        cand_ids = (
+            cluster.shard_indices(n_cand) if cluster is not None
+            else list(range(n_cand))
+        )
+        col_fold, col_class, col_lr, col_l2, col_model = [], [], [], [], []
+        local_tasks = []  # (task_id, model_idx)
+        model_idx = 0
+        for ci in cand_ids:
+            params = candidate_params[ci]
+            lr = float(params.get("lr", self.lr))
+            for fold in range(n_folds):
+                self_params = self.get_params()
+                merged = {**self_params, **params}
+                lam = _lam_from(merged, fold_train_n[fold])
+                for cc in cls:
+                    col_fold.append(fold)
+                    col_class.append(cc)
+                    col_lr.append(lr)
+                    col_l2.append(lam)
+                    col_model.append(model_idx)
+                local_tasks.append((ci * n_folds + fold, model_idx, fold))
+                model_idx += 1
+
+        results = {}
+        if model_idx > 0:
+            spec = ColumnSpec(
+                ds.device,
+                col_fold=np.asarray(col_fold, dtype=np.int32),
+                col_class=np.asarray(col_class, dtype=np.int32),
+                col_lr=np.asarray(col_lr, dtype=np.float32),
+                col_l2=np.asarray(col_l2, dtype=np.float32),
+            )
+            W = batched_sgd_fit(
+                ds, spec, self._loss, self.epochs, self.batch_size,
+                seed=self._seed(), momentum=self.momentum,
+            )
+            fit_time = time.perf_counter() - t0
+            t1 = time.perf_counter()
+            scores = batched_scores(
+                ds, spec, W, self._loss, np.asarray(col_model),
+                n_models=model_idx, n_classes=n_classes, metric=metric,
+            )
+            score_time = time.perf_counter() - t1
+            per = fit_time / max(model_idx, 1)
+            per_s = score_time / max(model_idx, 1)
+            for task_id, mi, fold in local_tasks:
+                results[task_id] = {
+                    "task_id": task_id,
+                    "test_scores": {"score": float(scores[mi])},
+                    "n_test": fold_test_n[fold],
+                    "fit_time": per,
+                    "score_time": per_s,
+                }
+        if cluster is not None:
+            out = cluster.gather_task_results(results, n_cand * n_folds)
+        else:
+            out = [results[i] for i in range(n_cand * n_folds)]
+        return out
+
+    def _device_metric(self, scoring):
+        if scoring is None:
+            metric = (
+                "accuracy" if isinstance(self, ClassifierMixin) else "r2"
+            )
+        elif isinstance(scoring, str):
+            metric = scoring
+        else:
+            raise FallbackToGeneric("callable/multi scoring not batched")
+        if metric not in _DEVICE_METRICS:
+            raise FallbackToGeneric(f"no device metric for {metric!r}")
+        return metric
+
+
+def _lam_from(params, n_train):
+    if "C" in params and params.get("C") is not None:
+        return 1.0 / (float(params["C"]) * max(n_train, 1))
+    return float(params.get("alpha", 1.0)) / max(n_train, 1)
+
+
+class LogisticRegression(ClassifierMixin, _BatchedLinearBase):
+    """Batched-SGD logistic regression (binary + internal one-vs-rest).
+
+    The solver differs from sklearn's lbfgs/liblinear (SURVEY.md §7 "exact
+    sklearn numerics"): features are standardized internally (coefficients
+    are mapped back to raw space, so the fitted model is exchangeable),
+    and the optimizer is mini-batch SGD — scores match sklearn to CV-noise
+    tolerance, not bitwise.
+    """
+
+    _loss = LOSS_LOG
+
+    def __init__(self, C=1.0, lr=0.5, epochs=20, batch_size=8192,
+                 momentum=0.9, standardize=True, random_state=None, sc=None):
+        self.C = C
+        self.lr = lr
+        self.epochs = epochs
+        self.batch_size = batch_size
+        self.momentum = momentum
+        self.standardize = standardize
+        self.random_state = random_state
+        self.sc = sc
+
+    def predict_proba(self, X):
+        scores = self.decision_function(X)
+        if scores.ndim == 1:
+            p = 1.0 / (1.0 + np.exp(-scores))
+            return np.column_stack([1.0 - p, p])
+        p = 1.0 / (1.0 + np.exp(-scores))
+        p /= p.sum(axis=1, keepdims=True)
+        return p
+
+    def predict_log_proba(self, X):
+        return np.log(np.clip(self.predict_proba(X), 1e-300, None))
+
+    def predict(self, X):
+        scores = self.decision_function(X)
+        if scores.ndim == 1:
+            return self.classes_[(scores >= 0).astype(int)]
+        return self.classes_[scores.argmax(axis=1)]
+
+
+class LinearSVC(ClassifierMixin, _BatchedLinearBase):
+    """Batched-SGD linear SVM (hinge loss, internal one-vs-rest)."""
+
+    _loss = LOSS_HINGE
+
+    def __init__(self, C=1.0, lr=0.5, epochs=20, batch_size=8192,
+                 momentum=0.9, standardize=True, random_state=None, sc=None):
+        self.C = C
+        self.lr = lr
+        self.epochs = epochs
+        self.batch_size = batch_size
+        self.momentum = momentum
+        self.standardize = standardize
+        self.random_state = random_state
+        self.sc = sc
+
+    def predict(self, X):
+        scores = self.decision_function(X)
+        if scores.ndim == 1:
+            return self.classes_[(scores >= 0).astype(int)]
+        return self.classes_[scores.argmax(axis=1)]
+
+
+class Ridge(RegressorMixin, _BatchedLinearBase):
+    """Batched-SGD ridge regression."""
+
+    _loss = LOSS_SQUARED
+
+    def __init__(self, alpha=1.0, lr=0.5, epochs=20, batch_size=8192,
+                 momentum=0.9, standardize=True, random_state=None, sc=None):
+        self.alpha = alpha
+        self.lr = lr
+        self.epochs = epochs
+        self.batch_size = batch_size
+        self.momentum = momentum
+        self.standardize = standardize
+        self.random_state = random_state
+        self.sc = sc
+
+    def predict(self, X):
+        return self.decision_function(X)

@@ -1,0 +1,199 @@
+"""
+Cluster — the ``sc`` analog: a single-node GPU task scheduler.
+
+sk-dist's distribution primitive is ``sc.broadcast(estimator)`` +
+``sc.parallelize(tasks).map(fit_one).collect()`` on Spark (reference:
+skdist/distribute/search.py:411-437 and SURVEY.md §2.3).  The MI355X-native
+replacement is SPMD over one process per GPU:
+
+  * ``torch.distributed`` with the nccl backend (RCCL over xGMI) when GPUs
+    are present; gloo for CPU-only multi-process tests;
+  * the training data is broadcast ONCE from rank 0 into every GPU's HBM
+    (288 GB each — no chunked-broadcast workaround needed, unlike Spark's
+    2 GB limit that forced multiclass.py:35-62's ``_split_X``);
+  * tasks are tiny descriptors (param dicts, fold ids, seeds, class ids)
+    sharded round-robin by task index — deterministic, never by completion
+    order (reference re-sorts after collect: search.py:439);
+  * results are all-gathered so every rank assembles the identical fitted
+    meta-estimator — no separate "driver refit + broadcast model" step.
+
+Outside torchrun the Cluster degrades to world_size==1 and simply marks
+"run tasks on the local GPU" — same code path, no communicator.
+
+A fitted estimator must never retain a Cluster (pickle contract, reference
+search.py:568-570): meta-estimators ``del self.sc`` before returning.
+"""
+
+import datetime
+import os
+
+import numpy as np
+import torch
+import torch.distributed as dist
+
+
+def _want_distributed():
+    return "RANK" in os.environ and "WORLD_SIZE" in os.environ
+
+
+class Cluster:
+    """Handle to the single-node GPU scheduler (the ``sc=`` argument).
+
+    Parameters
+    ----------
+    device : str or torch.device, optional
+        Compute device for this rank.  Defaults to ``cuda:<LOCAL_RANK>``
+        when GPUs are visible, else ``cpu`` (useful for multi-process
+        plumbing tests over gloo).
+    require_gpu : bool
+        If True, raise when no GPU is visible instead of falling back to
+        CPU — the bench path sets this so a silent eager fallback can
+        never produce a "GPU" number.
+    """
+
+    def __init__(self, device=None, require_gpu=False):
+        self._init_process_group_if_needed()
+        self.distributed = dist.is_available() and dist.is_initialized()
+        if self.distributed:
+            self.rank = dist.get_rank()
+            self.world_size = dist.get_world_size()
+        else:
+            self.rank = 0
+            self.world_size = 1
+        if device is not None:
+            self.device = torch.device(device)
+        elif torch.cuda.is_available():
+            local = int(os.environ.get("LOCAL_RANK", self.rank))
+            self.device = torch.device("cuda", local % torch.cuda.device_count())
+        else:
+            if require_gpu:
+                raise RuntimeError(
+                    "Cluster(require_gpu=True): no HIP device visible"
+                )
+            self.device = torch.device("cpu")
+        if self.device.type == "cuda":
+            torch.cuda.set_device(self.device)
+
+    @staticmethod
+    def _init_process_group_if_needed():
+        if not dist.is_available():
+            return
+        if dist.is_initialized() or not _want_distributed():
+            return
+        backend = "nccl" if torch.cuda.is_available() else "gloo"
+        dist.init_process_group(
+            backend=backend, timeout=datetime.timedelta(seconds=300)
+        )
+
+    # ------------------------------------------------------------------ #
+    # communication primitives
+    # ------------------------------------------------------------------ #
+
+    def barrier(self):
+        if self.distributed:
+            dist.barrier()
+
+    def bcast_obj(self, obj, src=0):
+        """Broadcast a small picklable object from ``src`` to all ranks."""
+        if not self.distributed:
+            return obj
+        box = [obj if self.rank == src else None]
+        dist.broadcast_object_list(box, src=src)
+        return box[0]
+
+    def bcast_tensor(self, t, src=0):
+        """Broadcast a tensor from ``src``; other ranks may pass None.
+
+        Shapes/dtypes are negotiated with a small object broadcast, then the
+        payload moves device-to-device over RCCL (xGMI p2p) in one shot.
+        Returns the tensor on ``self.device`` on every rank.
+        """
+        if not self.distributed:
+            return None if t is None else t.to(self.device)
+        meta = self.bcast_obj(
+            None if t is None else (tuple(t.shape), t.dtype), src=src
+        )
+        if meta is None:
+            return None
+        shape, dtype = meta
+        if self.rank == src:
+            t = t.to(self.device)
+        else:
+            t = torch.empty(shape, dtype=dtype, device=self.device)
+        dist.broadcast(t, src=src)
+        return t
+
+    def bcast_array(self, a, src=0, dtype=None):
+        """numpy → broadcast device tensor (the one-time (X, y) upload)."""
+        t = None
+        if self.rank == src and a is not None:
+            t = torch.as_tensor(np.ascontiguousarray(a))
+            if dtype is not None:
+                t = t.to(dtype)
+        return self.bcast_tensor(t, src=src)
+
+    def sync_host_data(self, *objs):
+        """Ensure every rank holds the same host-side data objects.
+
+        Callers in SPMD mode may pass data on rank 0 only; if any rank is
+        missing the first object, everything is broadcast from rank 0.
+        When all ranks already hold data (the common SPMD case) this is a
+        cheap flag exchange, no payload moves.
+        """
+        if not self.distributed:
+            return objs if len(objs) != 1 else objs[0]
+        flags = [None] * self.world_size
+        dist.all_gather_object(flags, objs[0] is None)
+        if any(flags):
+            objs = self.bcast_obj(objs, src=0)
+        return objs if len(objs) != 1 else objs[0]
+
+    # ------------------------------------------------------------------ #
+    # task fan-out
+    # ------------------------------------------------------------------ #
+
+    def shard_indices(self, n_tasks):
+        """Round-robin task-index shard for this rank (deterministic)."""
+        return list(range(self.rank, n_tasks, self.world_size))
+
+    def run_tasks(self, task_fn, tasks):
+        """Run ``task_fn(task)`` over all tasks, sharded across ranks.
+
+        Every rank receives the COMPLETE ordered result list (results keyed
+        by task index, so completion order never matters).  ``task_fn`` runs
+        on this rank's shard only; cross-rank results travel as pickled
+        objects (they are scores or compact fitted-model blobs — the data
+        itself never moves after the one-time broadcast).
+        """
+        mine = self.shard_indices(len(tasks))
+        local = {i: task_fn(tasks[i]) for i in mine}
+        return self.gather_task_results(local, len(tasks))
+
+    def gather_task_results(self, local, n_tasks):
+        """all-gather {task_id: result} dicts → full ordered list."""
+        if not self.distributed:
+            return [local[i] for i in range(n_tasks)]
+        boxes = [None] * self.world_size
+        dist.all_gather_object(boxes, local)
+        merged = {}
+        for d in boxes:
+            merged.update(d)
+        if len(merged) != n_tasks:
+            missing = sorted(set(range(n_tasks)) - set(merged))
+            raise RuntimeError(f"lost task results for ids {missing[:8]}")
+        return [merged[i] for i in range(n_tasks)]
+
+    # ------------------------------------------------------------------ #
+
+    def __repr__(self):
+        return (
+            f"Cluster(rank={self.rank}, world_size={self.world_size}, "
+            f"device={self.device})"
+        )
+
+    # A Cluster must never be pickled inside a fitted estimator.
+    def __reduce__(self):
+        raise TypeError(
+            "Cluster is not picklable: fitted estimators must strip `sc` "
+            "(sk-dist contract, reference search.py:568-570)"
+        )

@@ -1,0 +1,116 @@
+"""
+Tests for the GPU-native linear family (skdist_amd.models), run on the CPU
+torch path.  Quality parity targets come from the reference's published
+example outputs (BASELINE.md model-quality table).
+"""
+
+import pickle
+
+import numpy as np
+import pytest
+from sklearn.datasets import load_breast_cancer, load_digits, load_iris
+from sklearn.metrics import accuracy_score, r2_score, roc_auc_score
+from sklearn.model_selection import train_test_split
+
+from skdist_amd import Cluster
+from skdist_amd.distribute.search import DistGridSearchCV
+from skdist_amd.models import LinearSVC, LogisticRegression, Ridge
+
+
+def test_logreg_binary_quality():
+    X, y = load_breast_cancer(return_X_y=True)
+    clf = LogisticRegression(C=1.0, epochs=30, random_state=0)
+    clf.fit(X, y)
+    proba = clf.predict_proba(X)[:, 1]
+    auc = roc_auc_score(y, proba)
+    assert auc > 0.99, auc
+    acc = accuracy_score(y, clf.predict(X))
+    assert acc > 0.95, acc
+
+
+def test_logreg_multiclass():
+    X, y = load_digits(return_X_y=True)
+    X_tr, X_te, y_tr, y_te = train_test_split(
+        X, y, random_state=0, test_size=0.3
+    )
+    clf = LogisticRegression(epochs=30, random_state=0)
+    clf.fit(X_tr, y_tr)
+    acc = accuracy_score(y_te, clf.predict(X_te))
+    assert acc > 0.92, acc
+    p = clf.predict_proba(X_te)
+    assert p.shape == (len(y_te), 10)
+    assert np.allclose(p.sum(axis=1), 1.0, atol=1e-6)
+
+
+def test_linear_svc():
+    X, y = load_breast_cancer(return_X_y=True)
+    clf = LinearSVC(C=1.0, epochs=30, random_state=0)
+    clf.fit(X, y)
+    acc = accuracy_score(y, clf.predict(X))
+    assert acc > 0.95, acc
+
+
+def test_ridge():
+    rng = np.random.default_rng(0)
+    X = rng.normal(size=(500, 8))
+    w = rng.normal(size=8)
+    y = X @ w + 0.01 * rng.normal(size=500)
+    reg = Ridge(alpha=0.01, epochs=40, random_state=0)
+    reg.fit(X, y)
+    assert r2_score(y, reg.predict(X)) > 0.99
+
+
+def test_pickle_roundtrip():
+    X, y = load_iris(return_X_y=True)
+    clf = LogisticRegression(epochs=15, random_state=0).fit(X, y)
+    clf2 = pickle.loads(pickle.dumps(clf))
+    assert np.array_equal(clf.predict(X), clf2.predict(X))
+    # no device/scheduler state inside the pickle
+    assert not any(
+        "torch" in str(type(v)) for v in vars(clf2).values()
+    )
+
+
+def test_batched_grid_search_cpu_cluster():
+    """The batched device path through a world_size-1 CPU Cluster must give
+    the same ranking answer as the generic path."""
+    X, y = load_breast_cancer(return_X_y=True)
+    grid = {"C": [0.001, 0.1, 1.0]}
+    est = LogisticRegression(epochs=15, random_state=0)
+
+    gs_local = DistGridSearchCV(est, grid, cv=3, scoring="roc_auc")
+    gs_local.fit(X, y)
+
+    gs_batched = DistGridSearchCV(est, grid, cv=3, scoring="roc_auc",
+                                  sc=Cluster())
+    gs_batched.fit(X, y)
+
+    assert len(gs_batched.cv_results_["params"]) == 3
+    # batched scores agree with per-task generic scores to solver noise
+    a = gs_local.cv_results_["mean_test_score"]
+    b = gs_batched.cv_results_["mean_test_score"]
+    assert np.allclose(a, b, atol=0.02), (a, b)
+    assert gs_batched.best_score_ > 0.99
+    # result object pickles (sc stripped)
+    blob = pickle.dumps(gs_batched)
+    assert pickle.loads(blob).predict(X[:5]).shape == (5,)
+
+
+def test_batched_fallback_on_unsupported_scoring():
+    X, y = load_iris(return_X_y=True)
+    est = LogisticRegression(epochs=10, random_state=0)
+    gs = DistGridSearchCV(
+        est, {"C": [0.1, 1.0]}, cv=3,
+        scoring="balanced_accuracy",  # no device metric -> generic path
+        sc=Cluster(),
+    )
+    gs.fit(X, y)
+    assert gs.best_score_ > 0.8
+
+
+def test_batched_accuracy_default_scoring():
+    X, y = load_digits(return_X_y=True)
+    est = LogisticRegression(epochs=15, random_state=0)
+    gs = DistGridSearchCV(est, {"C": [0.1, 1.0]}, cv=3, sc=Cluster())
+    gs.fit(X, y)
+    assert gs.best_score_ > 0.9

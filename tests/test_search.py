@@ -1,0 +1,139 @@
+"""
+CPU-path tests for DistGridSearchCV / DistRandomizedSearchCV — mirrors the
+reference test strategy (skdist/distribute/tests/test_search.py and
+SURVEY.md §4): behavioral asserts on tiny data through the sc=None branch,
+plus pickle round-trips and cv_results_ schema checks.
+"""
+
+import pickle
+
+import numpy as np
+import pytest
+from scipy.stats import uniform
+from sklearn.datasets import load_iris
+from sklearn.linear_model import LogisticRegression as SkLogReg
+
+from skdist_amd.distribute.search import (
+    DistGridSearchCV,
+    DistRandomizedSearchCV,
+)
+
+
+@pytest.fixture
+def small_xy():
+    X = np.array([[1.0, 1.0, 1.0], [0.0, 0.0, 0.0], [-1.0, -1.0, -1.0]] * 100)
+    y = np.array([0, 0, 1] * 100)
+    return X, y
+
+
+def test_grid_search_local(small_xy):
+    X, y = small_xy
+    gs = DistGridSearchCV(
+        SkLogReg(solver="liblinear"), {"C": [0.1, 1.0, 10.0]}, cv=5
+    )
+    gs.fit(X, y)
+    preds = gs.predict(X[:3])
+    assert np.allclose(preds, [0, 0, 1])
+    assert gs.best_score_ > 0.9
+
+
+def test_randomized_search_local(small_xy):
+    X, y = small_xy
+    rs = DistRandomizedSearchCV(
+        SkLogReg(solver="liblinear"), {"C": uniform(0.1, 10)},
+        cv=5, n_iter=4, random_state=0,
+    )
+    rs.fit(X, y)
+    preds = rs.predict(X[:3])
+    assert np.allclose(preds, [0, 0, 1])
+
+
+def test_cv_results_schema(small_xy):
+    X, y = small_xy
+    gs = DistGridSearchCV(
+        SkLogReg(solver="liblinear"), {"C": [0.1, 1.0]}, cv=3
+    )
+    gs.fit(X, y)
+    r = gs.cv_results_
+    for key in (
+        "mean_fit_time", "std_fit_time", "mean_score_time", "std_score_time",
+        "params", "param_C", "mean_test_score", "std_test_score",
+        "rank_test_score", "split0_test_score", "split2_test_score",
+    ):
+        assert key in r, key
+    assert len(r["params"]) == 2
+    assert r["rank_test_score"].dtype == np.int32
+    # ranks: best candidate has rank 1
+    assert r["rank_test_score"].min() == 1
+
+
+def test_iris_baseline_config():
+    """BASELINE.json config 1: LogReg 4-point C grid, 3-fold, sc=None."""
+    X, y = load_iris(return_X_y=True)
+    gs = DistGridSearchCV(
+        SkLogReg(solver="liblinear"),
+        {"C": [0.01, 0.1, 1.0, 10.0]}, cv=3,
+    )
+    gs.fit(X, y)
+    assert gs.best_score_ > 0.9
+    assert len(gs.cv_results_["params"]) == 4
+
+
+def test_pickle_contract(small_xy):
+    """Fitted search strips sc and pickles (reference search.py:568-570)."""
+    X, y = small_xy
+    gs = DistGridSearchCV(SkLogReg(solver="liblinear"), {"C": [1.0]}, cv=3)
+    gs.fit(X, y)
+    assert gs.sc is None
+    blob = pickle.dumps(gs)
+    gs2 = pickle.loads(blob)
+    assert np.allclose(gs2.predict(X[:3]), [0, 0, 1])
+
+
+def test_multimetric(small_xy):
+    X, y = small_xy
+    gs = DistGridSearchCV(
+        SkLogReg(solver="liblinear"), {"C": [0.1, 1.0]},
+        scoring=["accuracy", "roc_auc"], refit="roc_auc", cv=3,
+    )
+    gs.fit(X, y)
+    assert "mean_test_roc_auc" in gs.cv_results_
+    assert "mean_test_accuracy" in gs.cv_results_
+    assert gs.best_score_ > 0.9
+
+
+def test_error_score(small_xy):
+    X, y = small_xy
+
+    class Broken(SkLogReg):
+        def fit(self, *a, **k):
+            raise RuntimeError("boom")
+
+    gs = DistGridSearchCV(
+        Broken(), {"C": [1.0]}, cv=3, error_score=0.0, refit=False
+    )
+    with pytest.warns(Warning):
+        gs.fit(X, y)
+    assert np.allclose(gs.cv_results_["mean_test_score"], 0.0)
+
+    gs = DistGridSearchCV(Broken(), {"C": [1.0]}, cv=3, error_score="raise")
+    with pytest.raises(RuntimeError):
+        gs.fit(X, y)
+
+
+def test_n_jobs_parallel(small_xy):
+    X, y = small_xy
+    gs = DistGridSearchCV(
+        SkLogReg(solver="liblinear"), {"C": [0.1, 1.0]}, cv=3, n_jobs=2
+    )
+    gs.fit(X, y)
+    assert gs.best_score_ > 0.9
+
+
+def test_preds_attribute(small_xy):
+    X, y = small_xy
+    gs = DistGridSearchCV(
+        SkLogReg(solver="liblinear"), {"C": [1.0]}, cv=3, preds=True
+    )
+    gs.fit(X, y)
+    assert gs.preds_.shape[0] == len(y)
