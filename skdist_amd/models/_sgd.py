@@ -131,8 +131,19 @@ class DeviceDataset:
         else:
             self.feat_mean = np.zeros(self.f, dtype=np.float32)
             self.feat_std = np.ones(self.f, dtype=np.float32)
-        ones = torch.ones(self.n, 1, dtype=Xt.dtype, device=self.device)
-        self.Xaug = torch.cat([Xt, ones], dim=1).to(comp_dtype).contiguous()
+        # [X | 1 | 0-pad]: the ones column is the intercept; on GPU the
+        # width is padded to a multiple of 32 for the MFMA K-loop (pad
+        # features are zero, so their weights stay exactly 0).
+        self.intercept_row = self.f
+        fa = self.f + 1
+        pad = (-fa) % 32 if self.device.type == "cuda" else 0
+        self.fa = fa + pad
+        cols = [Xt, torch.ones(self.n, 1, dtype=Xt.dtype, device=self.device)]
+        if pad:
+            cols.append(
+                torch.zeros(self.n, pad, dtype=Xt.dtype, device=self.device)
+            )
+        self.Xaug = torch.cat(cols, dim=1).to(comp_dtype).contiguous()
         del Xt
 
         self.fold_id = None  # set by set_cv_partition
@@ -188,17 +199,18 @@ def batched_sgd_fit(ds, spec, loss, epochs, batch_size, seed=0,
     """
     device = ds.device
     n, fa = ds.Xaug.shape
-    W = torch.zeros(fa, spec.ncols, dtype=torch.float32, device=device)
-    V = (
-        torch.zeros_like(W) if momentum > 0.0 else None
-    )
-    hip = (not force_eager) and _use_hip(device)
-
-    rng = np.random.default_rng(seed)
     loss_id = _LOSS_IDS[loss] if isinstance(loss, str) else loss
+    hip = (not force_eager) and _use_hip(device)
+    if hip:
+        from ..ops import hip_sgd_solve
 
-    target_base = ds.y_float if ds.y_int is None else None
+        return hip_sgd_solve(
+            ds, spec, loss_id, epochs, batch_size, seed, momentum, lr_decay
+        )
 
+    W = torch.zeros(fa, spec.ncols, dtype=torch.float32, device=device)
+    V = torch.zeros_like(W) if momentum > 0.0 else None
+    rng = np.random.default_rng(seed)
     for epoch in range(epochs):
         perm = torch.as_tensor(
             rng.permutation(n), dtype=torch.int64, device=device
@@ -206,23 +218,15 @@ def batched_sgd_fit(ds, spec, loss, epochs, batch_size, seed=0,
         lr_scale = 1.0 / (1.0 + lr_decay * epoch)
         for start in range(0, n, batch_size):
             idx = perm[start : start + batch_size]
-            if hip:
-                from ..ops import sgd_step_hip
-
-                sgd_step_hip(
-                    ds.Xaug, ds.y_float, ds.fold_id, idx, W, V, spec,
-                    loss_id, lr_scale, momentum,
-                )
-            else:
-                _sgd_step_torch(
-                    ds.Xaug, ds.y_float, ds.fold_id, idx, W, V, spec,
-                    loss_id, lr_scale, momentum,
-                )
+            _sgd_step_torch(
+                ds.Xaug, ds.y_float, ds.fold_id, idx, W, V, spec,
+                loss_id, lr_scale, momentum, ds.intercept_row,
+            )
     return W
 
 
 def _sgd_step_torch(Xaug, y_float, fold_id, idx, W, V, spec, loss_id,
-                    lr_scale, momentum):
+                    lr_scale, momentum, intercept_row=None):
     """One mini-batch step, eager torch — the numerics reference the HIP
     kernels are tested against (fp32 on CPU; on GPU it mirrors the kernel's
     bf16-in/fp32-accumulate)."""
@@ -254,7 +258,10 @@ def _sgd_step_torch(Xaug, y_float, fold_id, idx, W, V, spec, loss_id,
 
     grad = (Xb.transpose(0, 1).to(comp) @ G.to(comp)).to(torch.float32) / m
     # L2 on weights only, not the intercept row
-    grad[:-1] += spec.col_l2.unsqueeze(0) * W[:-1]
+    ir = W.shape[0] - 1 if intercept_row is None else intercept_row
+    l2 = spec.col_l2.unsqueeze(0) * W
+    l2[ir] = 0.0
+    grad += l2
     step = spec.col_lr.unsqueeze(0) * lr_scale * grad
     if momentum > 0.0:
         V.mul_(momentum).add_(step)

@@ -100,9 +100,9 @@ class _BatchedLinearBase(BaseEstimator):
         return 0 if rs is None else int(rs)
 
     def _store_fitted(self, ds, W, is_clf):
-        Wh = W.cpu().numpy()  # [f+1, ncols]
-        w_std = Wh[:-1].T     # [ncols, f] standardized space
-        b_std = Wh[-1]
+        Wh = W.cpu().numpy()       # [fa(+pad), ncols]
+        w_std = Wh[: ds.f].T       # [ncols, f] standardized space
+        b_std = Wh[ds.intercept_row]
         w_raw = np.empty_like(w_std)
         b_raw = np.empty_like(b_std)
         for c in range(w_std.shape[0]):

@@ -46,13 +46,63 @@ def require_hip():
     return _ext
 
 
-def sgd_step_hip(Xaug, y_float, fold_id, idx, W, V, spec, loss_id,
-                 lr_scale, momentum):
-    """One fused mini-batch SGD step on the HIP kernels."""
+def _pad_cols(t, ncols_pad, fill):
+    import torch as _t
+
+    out = _t.full((ncols_pad,), fill, dtype=t.dtype, device=t.device)
+    out[: t.shape[0]] = t
+    return out.contiguous()
+
+
+def hip_sgd_solve(ds, spec, loss_id, epochs, batch_size, seed, momentum,
+                  lr_decay, splitk=8):
+    """Full batched SGD solve on the HIP kernels (K1+K2+K3 per step).
+
+    Walks the same per-epoch host-RNG permutations as the torch reference
+    (minibatch = slab of the epoch-shuffled copy), so the two paths are
+    comparable to bf16 tolerance.  Returns W [fa, ncols] fp32.
+    """
+    import numpy as np
+
     ext = require_hip()
-    ext.sgd_step(
-        Xaug, y_float, fold_id, idx, W,
-        V if V is not None else torch.empty(0, device=W.device),
-        spec.col_fold, spec.col_class, spec.col_lr, spec.col_l2,
-        int(loss_id), float(lr_scale), float(momentum),
+    device = ds.device
+    n, fa = ds.Xaug.shape
+    assert fa % 32 == 0
+    ncols = spec.ncols
+    ncp = (ncols + 127) // 128 * 128
+    bs = min(batch_size, n)
+    gts = (bs + 127) // 128 * 128
+
+    W = torch.zeros(fa, ncp, dtype=torch.float32, device=device)
+    V = (
+        torch.zeros_like(W) if momentum > 0.0
+        else torch.empty(0, device=device)
     )
+    WbfT = torch.zeros(ncp, fa, dtype=torch.bfloat16, device=device)
+    GT = torch.empty(ncp, gts, dtype=torch.bfloat16, device=device)
+    partial = torch.empty(splitk, fa, ncp, dtype=torch.float32,
+                          device=device)
+    cls_p = _pad_cols(spec.col_class, ncp, -99)
+    cfold_p = _pad_cols(spec.col_fold, ncp, -9)
+    lr_p = _pad_cols(spec.col_lr, ncp, 0.0)
+    l2_p = _pad_cols(spec.col_l2, ncp, 0.0)
+
+    rng = np.random.default_rng(seed)
+    for epoch in range(epochs):
+        perm = torch.as_tensor(
+            rng.permutation(n), dtype=torch.int64, device=device
+        )
+        Xs = ds.Xaug.index_select(0, perm).contiguous()
+        XsT = Xs.t().contiguous()
+        ys = ds.y_float.index_select(0, perm).contiguous()
+        folds = ds.fold_id.index_select(0, perm).to(torch.int32).contiguous()
+        lr_scale = 1.0 / (1.0 + lr_decay * epoch)
+        for start in range(0, n, bs):
+            m = min(bs, n - start)
+            ext.sgd_step(
+                Xs, XsT, GT, W, V, WbfT, partial, ys, folds,
+                cls_p, cfold_p, lr_p, l2_p,
+                start, m, int(loss_id), float(lr_scale), float(momentum),
+                int(ds.intercept_row),
+            )
+    return W[:, :ncols]

@@ -1,0 +1,66 @@
+// Python bindings for the skdist_amd HIP kernels (built with hipcc against
+// the ROCm torch in this image; no CUDA shims, no hipify).
+#include <torch/extension.h>
+#include <c10/hip/HIPStream.h>
+
+#include <hip/hip_runtime.h>
+
+extern "C" hipError_t skdist_sgd_step(
+    const void* Xs, const void* XsT, const void* WbfT_in,
+    void* GT, void* W, void* V, void* WbfT, void* partial,
+    const void* y, const void* fold,
+    const void* col_class, const void* col_fold,
+    const void* col_lr, const void* col_l2,
+    long long start, long long m, long long n, int fa, int ncols_pad,
+    int gt_stride, int splitk, int loss_id,
+    float lr_scale, float momentum, int intercept_row,
+    hipStream_t stream);
+
+namespace {
+
+#define CHECK_DEV(t) TORCH_CHECK((t).is_cuda(), #t " must be on the GPU")
+#define CHECK_CONT(t) TORCH_CHECK((t).is_contiguous(), #t " not contiguous")
+
+void sgd_step(torch::Tensor Xs, torch::Tensor XsT, torch::Tensor GT,
+              torch::Tensor W, torch::Tensor V, torch::Tensor WbfT,
+              torch::Tensor partial, torch::Tensor y, torch::Tensor fold,
+              torch::Tensor col_class, torch::Tensor col_fold,
+              torch::Tensor col_lr, torch::Tensor col_l2,
+              int64_t start, int64_t m, int64_t loss_id, double lr_scale,
+              double momentum, int64_t intercept_row) {
+    for (auto* t : {&Xs, &XsT, &GT, &W, &WbfT, &partial, &y, &fold,
+                    &col_class, &col_fold, &col_lr, &col_l2}) {
+        CHECK_DEV(*t);
+        CHECK_CONT(*t);
+    }
+    const auto n = Xs.size(0);
+    const auto fa = Xs.size(1);
+    const auto ncols_pad = W.size(1);
+    const auto gt_stride = GT.size(1);
+    const auto splitk = partial.size(0);
+    TORCH_CHECK(fa % 32 == 0, "fa must be a multiple of 32");
+    TORCH_CHECK(ncols_pad % 128 == 0, "ncols_pad must be a multiple of 128");
+    TORCH_CHECK(Xs.scalar_type() == torch::kBFloat16, "Xs must be bf16");
+    TORCH_CHECK(XsT.size(0) == fa && XsT.size(1) == n, "XsT shape");
+    TORCH_CHECK((n * fa) * 2 < (1LL << 31),
+                "X too large for 32-bit buffer offsets; shard rows");
+    const bool has_V = V.numel() > 0;
+
+    auto stream = c10::hip::getCurrentHIPStream().stream();
+    hipError_t err = skdist_sgd_step(
+        Xs.data_ptr(), XsT.data_ptr(), WbfT.data_ptr(), GT.data_ptr(),
+        W.data_ptr(), has_V ? V.data_ptr() : nullptr, WbfT.data_ptr(),
+        partial.data_ptr(), y.data_ptr(), fold.data_ptr(),
+        col_class.data_ptr(), col_fold.data_ptr(), col_lr.data_ptr(),
+        col_l2.data_ptr(), start, m, n, (int)fa, (int)ncols_pad,
+        (int)gt_stride, (int)splitk, (int)loss_id, (float)lr_scale,
+        (float)momentum, (int)intercept_row, stream);
+    TORCH_CHECK(err == hipSuccess, "skdist_sgd_step: ",
+                hipGetErrorString(err));
+}
+
+}  // namespace
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+    m.def("sgd_step", &sgd_step, "fused batched SGD step (K1+K2+K3)");
+}
