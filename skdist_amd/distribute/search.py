@@ -163,6 +163,7 @@ class DistBaseSearchCV(BaseEstimator, MetaEstimatorMixin, metaclass=ABCMeta):
             and not fit_params
         )
         out = None
+        refit_models = None
         if batched:
             from ..models.linear import FallbackToGeneric
 
@@ -178,6 +179,9 @@ class DistBaseSearchCV(BaseEstimator, MetaEstimatorMixin, metaclass=ABCMeta):
                 )
             except FallbackToGeneric:
                 out = None
+        if isinstance(out, dict):
+            refit_models = out.get("refit_estimators")
+            out = out["tasks"]
         if out is None:
             out = self._run_task_grid(
                 base_estimator, X, y, scorers, candidate_params, cv_splits,
@@ -200,18 +204,26 @@ class DistBaseSearchCV(BaseEstimator, MetaEstimatorMixin, metaclass=ABCMeta):
                 results[f"mean_test_{refit_metric}"][self.best_index_]
             )
         if self.refit:
-            best = _clone(base_estimator)
-            best.set_params(**self.best_params_)
-            if hasattr(best, "sc"):
-                best.sc = sc  # refit may use the GPU, stripped right after
-            start = time.perf_counter()
-            if y is not None:
-                best.fit(X, y, **fit_params)
+            if refit_models is not None and self.best_index_ in refit_models:
+                # batched path: the full-data model trained alongside the
+                # CV columns — refit is a column extraction, not a 2nd solve
+                self.best_estimator_ = refit_models[self.best_index_]
+                self.refit_time_ = getattr(
+                    self.best_estimator_, "fit_time_", 0.0
+                )
             else:
-                best.fit(X, **fit_params)
-            self.refit_time_ = time.perf_counter() - start
-            _strip_sc(best)
-            self.best_estimator_ = best
+                best = _clone(base_estimator)
+                best.set_params(**self.best_params_)
+                if hasattr(best, "sc"):
+                    best.sc = sc  # may use the GPU, stripped right after
+                start = time.perf_counter()
+                if y is not None:
+                    best.fit(X, y, **fit_params)
+                else:
+                    best.fit(X, **fit_params)
+                self.refit_time_ = time.perf_counter() - start
+                _strip_sc(best)
+                self.best_estimator_ = best
             if self.preds:
                 self.preds_ = self._out_of_fold_preds(
                     base_estimator, X, y, cv_splits, fit_params

@@ -169,6 +169,22 @@ class DeviceDataset:
         self.fold_id = torch.as_tensor(fold, device=self.device)
         return True
 
+    def shuffled_views(self, perm):
+        """Row-shuffled (Xs, XsT, y, fold) device copies (cached: every
+        solve in a search shares the same seed and thus the same views)."""
+        key = (int(perm[0]), int(perm[-1]), len(perm))
+        if getattr(self, "_shuf_key", None) == key:
+            return self._shuf
+        Xs = self.Xaug.index_select(0, perm).contiguous()
+        XsT = Xs.t().contiguous()
+        ys = self.y_float.index_select(0, perm).contiguous()
+        folds = (
+            self.fold_id.index_select(0, perm).to(torch.int32).contiguous()
+        )
+        self._shuf = (Xs, XsT, ys, folds)
+        self._shuf_key = key
+        return self._shuf
+
     def unstandardize_coef(self, w, b):
         """Map standardized-space (w, b) back to raw-feature space."""
         w_raw = w / self.feat_std
@@ -211,10 +227,10 @@ def batched_sgd_fit(ds, spec, loss, epochs, batch_size, seed=0,
     W = torch.zeros(fa, spec.ncols, dtype=torch.float32, device=device)
     V = torch.zeros_like(W) if momentum > 0.0 else None
     rng = np.random.default_rng(seed)
+    perm = torch.as_tensor(
+        rng.permutation(n), dtype=torch.int64, device=device
+    )
     for epoch in range(epochs):
-        perm = torch.as_tensor(
-            rng.permutation(n), dtype=torch.int64, device=device
-        )
         lr_scale = 1.0 / (1.0 + lr_decay * epoch)
         for start in range(0, n, batch_size):
             idx = perm[start : start + batch_size]
@@ -321,7 +337,11 @@ class _MetricState:
         self.device = device
         z = lambda *shape: torch.zeros(*shape, dtype=torch.float64,
                                        device=device)
-        if metric in ("accuracy", "f1", "f1_weighted", "f1_macro"):
+        if metric == "accuracy":
+            # sum-based fast path: no scatter, no nonzero
+            self.correct = z(n_models)
+            self.count = z(n_models)
+        elif metric in ("f1", "f1_weighted", "f1_macro"):
             k = self.n_classes
             self.confusion = z(n_models, k, k)  # [model, true, pred]
         elif metric == "neg_log_loss":
@@ -350,7 +370,12 @@ class _MetricState:
         else:
             pred = (Z >= 0).to(torch.int64)                # [m, nm]
 
-        if self.metric in ("accuracy", "f1", "f1_weighted", "f1_macro"):
+        if self.metric == "accuracy":
+            true = yb.to(torch.int64).unsqueeze(1)
+            hit = (pred == true) & tmask
+            self.correct += hit.sum(dim=0).to(torch.float64)
+            self.count += tmask.sum(dim=0).to(torch.float64)
+        elif self.metric in ("f1", "f1_weighted", "f1_macro"):
             k = self.n_classes
             true = yb.to(torch.int64).unsqueeze(1).expand(m, nm)
             flat = (
@@ -401,10 +426,7 @@ class _MetricState:
 
     def finalize(self):
         if self.metric == "accuracy":
-            conf = self.confusion
-            correct = conf.diagonal(dim1=1, dim2=2).sum(dim=1)
-            total = conf.sum(dim=(1, 2)).clamp_min(1)
-            return (correct / total).cpu().numpy()
+            return (self.correct / self.count.clamp_min(1)).cpu().numpy()
         if self.metric in ("f1", "f1_weighted", "f1_macro"):
             conf = self.confusion
             tp = conf.diagonal(dim1=1, dim2=2)              # [nm, k]

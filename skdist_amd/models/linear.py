@@ -180,14 +180,15 @@ class _BatchedLinearBase(BaseEstimator):
             else list(range(n_cand))
         )
         col_fold, col_class, col_lr, col_l2, col_model = [], [], [], [], []
-        local_tasks = []  # (task_id, model_idx)
+        local_tasks = []   # (task_id, model_idx, fold)
+        full_models = {}   # cand_id -> model_idx of the full-data model
         model_idx = 0
+        self_params = self.get_params()
         for ci in cand_ids:
             params = candidate_params[ci]
             lr = float(params.get("lr", self.lr))
+            merged = {**self_params, **params}
             for fold in range(n_folds):
-                self_params = self.get_params()
-                merged = {**self_params, **params}
                 lam = _lam_from(merged, fold_train_n[fold])
                 for cc in cls:
                     col_fold.append(fold)
@@ -197,8 +198,22 @@ class _BatchedLinearBase(BaseEstimator):
                     col_model.append(model_idx)
                 local_tasks.append((ci * n_folds + fold, model_idx, fold))
                 model_idx += 1
+            # one extra full-data model per candidate: trains alongside the
+            # fold models so the search's "refit" is a column extraction
+            # instead of a second solve (MI355X-batched replacement for
+            # reference search.py:543-550's driver-side refit)
+            lam = _lam_from(merged, ds.n)
+            for cc in cls:
+                col_fold.append(-2)  # trains on every row
+                col_class.append(cc)
+                col_lr.append(lr)
+                col_l2.append(lam)
+                col_model.append(model_idx)
+            full_models[ci] = model_idx
+            model_idx += 1
 
         results = {}
+        refit_local = {}
         if model_idx > 0:
             spec = ColumnSpec(
                 ds.device,
@@ -228,11 +243,34 @@ class _BatchedLinearBase(BaseEstimator):
                     "fit_time": per,
                     "score_time": per_s,
                 }
+            # materialize a fitted estimator per candidate from the
+            # full-data columns (host numpy weights, sc-free, picklable)
+            ncols_per = len(cls)
+            Wh = W
+            for ci, mi in full_models.items():
+                cols = slice(mi * ncols_per, (mi + 1) * ncols_per)
+                est = sk_clone_without_sc(self)
+                est.set_params(**candidate_params[ci])
+                est._store_fitted(ds, Wh[:, cols], is_clf)
+                est.n_features_in_ = ds.f
+                est.fit_time_ = per
+                refit_local[ci] = est
         if cluster is not None:
             out = cluster.gather_task_results(results, n_cand * n_folds)
+            boxes = [None] * cluster.world_size
+            if cluster.distributed:
+                import torch.distributed as dist
+
+                dist.all_gather_object(boxes, refit_local)
+                refit_all = {}
+                for b in boxes:
+                    refit_all.update(b)
+            else:
+                refit_all = refit_local
         else:
             out = [results[i] for i in range(n_cand * n_folds)]
-        return out
+            refit_all = refit_local
+        return {"tasks": out, "refit_estimators": refit_all}
 
     def _device_metric(self, scoring):
         if scoring is None:
@@ -252,6 +290,19 @@ def _lam_from(params, n_train):
     if "C" in params and params.get("C") is not None:
         return 1.0 / (float(params["C"]) * max(n_train, 1))
     return float(params.get("alpha", 1.0)) / max(n_train, 1)
+
+
+def sk_clone_without_sc(est):
+    """Unfitted copy of est with sc=None (for batch-materialized models)."""
+    from sklearn.base import clone
+
+    sc = getattr(est, "sc", None)
+    est.sc = None
+    try:
+        out = clone(est)
+    finally:
+        est.sc = sc
+    return out
 
 
 class LogisticRegression(ClassifierMixin, _BatchedLinearBase):

@@ -87,22 +87,19 @@ def hip_sgd_solve(ds, spec, loss_id, epochs, batch_size, seed, momentum,
     lr_p = _pad_cols(spec.col_lr, ncp, 0.0)
     l2_p = _pad_cols(spec.col_l2, ncp, 0.0)
 
+    # one seeded shuffle (minibatch composition then stays fixed across
+    # epochs — standard for convex SGD; matches the torch reference path)
     rng = np.random.default_rng(seed)
+    perm = torch.as_tensor(
+        rng.permutation(n), dtype=torch.int64, device=device
+    )
+    Xs, XsT, ys, folds = ds.shuffled_views(perm)
     for epoch in range(epochs):
-        perm = torch.as_tensor(
-            rng.permutation(n), dtype=torch.int64, device=device
-        )
-        Xs = ds.Xaug.index_select(0, perm).contiguous()
-        XsT = Xs.t().contiguous()
-        ys = ds.y_float.index_select(0, perm).contiguous()
-        folds = ds.fold_id.index_select(0, perm).to(torch.int32).contiguous()
         lr_scale = 1.0 / (1.0 + lr_decay * epoch)
-        for start in range(0, n, bs):
-            m = min(bs, n - start)
-            ext.sgd_step(
-                Xs, XsT, GT, W, V, WbfT, partial, ys, folds,
-                cls_p, cfold_p, lr_p, l2_p,
-                start, m, int(loss_id), float(lr_scale), float(momentum),
-                int(ds.intercept_row),
-            )
+        ext.sgd_epoch(
+            Xs, XsT, GT, W, V, WbfT, partial, ys, folds,
+            cls_p, cfold_p, lr_p, l2_p,
+            bs, int(loss_id), float(lr_scale), float(momentum),
+            int(ds.intercept_row),
+        )
     return W[:, :ncols]

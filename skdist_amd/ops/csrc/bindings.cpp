@@ -59,8 +59,39 @@ void sgd_step(torch::Tensor Xs, torch::Tensor XsT, torch::Tensor GT,
                 hipGetErrorString(err));
 }
 
+// one whole epoch: the minibatch loop runs in C++ so the hot path costs
+// one pybind crossing per epoch instead of one per step
+void sgd_epoch(torch::Tensor Xs, torch::Tensor XsT, torch::Tensor GT,
+               torch::Tensor W, torch::Tensor V, torch::Tensor WbfT,
+               torch::Tensor partial, torch::Tensor y, torch::Tensor fold,
+               torch::Tensor col_class, torch::Tensor col_fold,
+               torch::Tensor col_lr, torch::Tensor col_l2,
+               int64_t batch_size, int64_t loss_id, double lr_scale,
+               double momentum, int64_t intercept_row) {
+    const auto n = Xs.size(0);
+    const auto fa = Xs.size(1);
+    const auto ncols_pad = W.size(1);
+    const auto gt_stride = GT.size(1);
+    const auto splitk = partial.size(0);
+    const bool has_V = V.numel() > 0;
+    auto stream = c10::hip::getCurrentHIPStream().stream();
+    for (int64_t start = 0; start < n; start += batch_size) {
+        const int64_t m = std::min(batch_size, n - start);
+        hipError_t err = skdist_sgd_step(
+            Xs.data_ptr(), XsT.data_ptr(), WbfT.data_ptr(), GT.data_ptr(),
+            W.data_ptr(), has_V ? V.data_ptr() : nullptr, WbfT.data_ptr(),
+            partial.data_ptr(), y.data_ptr(), fold.data_ptr(),
+            col_class.data_ptr(), col_fold.data_ptr(), col_lr.data_ptr(),
+            col_l2.data_ptr(), start, m, n, (int)fa, (int)ncols_pad,
+            (int)gt_stride, (int)splitk, (int)loss_id, (float)lr_scale,
+            (float)momentum, (int)intercept_row, stream);
+        TORCH_CHECK(err == hipSuccess, "sgd_epoch: ", hipGetErrorString(err));
+    }
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("sgd_step", &sgd_step, "fused batched SGD step (K1+K2+K3)");
+    m.def("sgd_epoch", &sgd_epoch, "one epoch of fused SGD steps");
 }

@@ -24,9 +24,9 @@ def _ref_solve(Xf32, y, fold, col_class, col_fold, col_lr, col_l2,
     V = torch.zeros_like(W)
     rng = np.random.default_rng(seed)
     Xb = _bf16_round(Xf32)
+    perm = rng.permutation(n)  # one shuffle, fixed minibatches (as solver)
+    Xs, ys, fs = Xb[perm], y[perm], fold[perm]
     for _ in range(epochs):
-        perm = rng.permutation(n)
-        Xs, ys, fs = Xb[perm], y[perm], fold[perm]
         for s in range(0, n, bs):
             Xm, ym, fm = Xs[s:s+bs], ys[s:s+bs], fs[s:s+bs]
             m = Xm.shape[0]
