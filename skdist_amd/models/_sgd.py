@@ -112,12 +112,14 @@ class DeviceDataset:
             self.y_int = None
         else:
             y_np = yt.cpu().numpy()
-            self.classes_ = (
-                np.asarray(classes) if classes is not None else np.unique(y_np)
+            if classes is not None:
+                self.classes_ = np.asarray(classes)
+                enc = np.searchsorted(self.classes_, y_np)
+            else:
+                self.classes_, enc = np.unique(y_np, return_inverse=True)
+            self.y_int = torch.as_tensor(
+                np.ascontiguousarray(enc, dtype=np.int32), device=self.device
             )
-            lut = {v: i for i, v in enumerate(self.classes_)}
-            enc = np.asarray([lut[v] for v in y_np], dtype=np.int32)
-            self.y_int = torch.as_tensor(enc, device=self.device)
             self.y_float = self.y_int.to(torch.float32)
 
         # standardize + augment with ones column (intercept)
@@ -170,13 +172,22 @@ class DeviceDataset:
         return True
 
     def shuffled_views(self, perm):
-        """Row-shuffled (Xs, XsT, y, fold) device copies (cached: every
-        solve in a search shares the same seed and thus the same views)."""
+        """Row-shuffled (Xs, XsT, y, fold) device copies, padded for the
+        HIP kernels: Xs to n_pad rows (mult of 128, zero tail), XsT to
+        fa_store rows (mult of 128).  Cached — every solve in a search
+        shares the same seed and thus the same views."""
         key = (int(perm[0]), int(perm[-1]), len(perm))
         if getattr(self, "_shuf_key", None) == key:
             return self._shuf
-        Xs = self.Xaug.index_select(0, perm).contiguous()
-        XsT = Xs.t().contiguous()
+        n, fa = self.Xaug.shape
+        n_pad = (n + 127) // 128 * 128
+        fa_store = (fa + 127) // 128 * 128
+        Xs = torch.zeros(n_pad, fa, dtype=self.Xaug.dtype,
+                         device=self.device)
+        torch.index_select(self.Xaug, 0, perm, out=Xs[:n])
+        XsT = torch.zeros(fa_store, n_pad, dtype=self.Xaug.dtype,
+                          device=self.device)
+        XsT[:fa] = Xs.t()
         ys = self.y_float.index_select(0, perm).contiguous()
         folds = (
             self.fold_id.index_select(0, perm).to(torch.int32).contiguous()

@@ -70,7 +70,8 @@ def hip_sgd_solve(ds, spec, loss_id, epochs, batch_size, seed, momentum,
     assert fa % 32 == 0
     ncols = spec.ncols
     ncp = (ncols + 127) // 128 * 128
-    bs = min(batch_size, n)
+    bs = min(batch_size, (n + 127) // 128 * 128)
+    bs = max(128, bs - bs % 128)
     gts = (bs + 127) // 128 * 128
 
     W = torch.zeros(fa, ncp, dtype=torch.float32, device=device)

@@ -11,7 +11,8 @@ extern "C" hipError_t skdist_sgd_step(
     const void* y, const void* fold,
     const void* col_class, const void* col_fold,
     const void* col_lr, const void* col_l2,
-    long long start, long long m, long long n, int fa, int ncols_pad,
+    long long start, long long m, long long n, long long n_pad,
+    long long fa_store, int fa, int ncols_pad,
     int gt_stride, int splitk, int loss_id,
     float lr_scale, float momentum, int intercept_row,
     hipStream_t stream);
@@ -20,6 +21,31 @@ namespace {
 
 #define CHECK_DEV(t) TORCH_CHECK((t).is_cuda(), #t " must be on the GPU")
 #define CHECK_CONT(t) TORCH_CHECK((t).is_contiguous(), #t " not contiguous")
+
+struct StepArgs {
+    int64_t n_pad, fa, fa_store, ncols_pad, gt_stride, splitk, n;
+};
+
+StepArgs check_args(torch::Tensor& Xs, torch::Tensor& XsT,
+                    torch::Tensor& GT, torch::Tensor& W,
+                    torch::Tensor& partial, torch::Tensor& y) {
+    StepArgs a;
+    a.n_pad = Xs.size(0);
+    a.fa = Xs.size(1);
+    a.fa_store = XsT.size(0);
+    a.ncols_pad = W.size(1);
+    a.gt_stride = GT.size(1);
+    a.splitk = partial.size(0);
+    a.n = y.size(0);
+    TORCH_CHECK(a.fa % 32 == 0, "fa must be a multiple of 32");
+    TORCH_CHECK(a.n_pad % 128 == 0, "n_pad must be a multiple of 128");
+    TORCH_CHECK(a.fa_store % 128 == 0, "fa_store must be a multiple of 128");
+    TORCH_CHECK(a.ncols_pad % 128 == 0, "ncols_pad must be a mult of 128");
+    TORCH_CHECK(Xs.scalar_type() == torch::kBFloat16, "Xs must be bf16");
+    TORCH_CHECK(XsT.size(1) == a.n_pad, "XsT shape mismatch");
+    TORCH_CHECK(W.size(0) == a.fa, "W rows != fa");
+    return a;
+}
 
 void sgd_step(torch::Tensor Xs, torch::Tensor XsT, torch::Tensor GT,
               torch::Tensor W, torch::Tensor V, torch::Tensor WbfT,
@@ -33,28 +59,17 @@ void sgd_step(torch::Tensor Xs, torch::Tensor XsT, torch::Tensor GT,
         CHECK_DEV(*t);
         CHECK_CONT(*t);
     }
-    const auto n = Xs.size(0);
-    const auto fa = Xs.size(1);
-    const auto ncols_pad = W.size(1);
-    const auto gt_stride = GT.size(1);
-    const auto splitk = partial.size(0);
-    TORCH_CHECK(fa % 32 == 0, "fa must be a multiple of 32");
-    TORCH_CHECK(ncols_pad % 128 == 0, "ncols_pad must be a multiple of 128");
-    TORCH_CHECK(Xs.scalar_type() == torch::kBFloat16, "Xs must be bf16");
-    TORCH_CHECK(XsT.size(0) == fa && XsT.size(1) == n, "XsT shape");
-    TORCH_CHECK((n * fa) * 2 < (1LL << 31),
-                "X too large for 32-bit buffer offsets; shard rows");
+    auto a = check_args(Xs, XsT, GT, W, partial, y);
     const bool has_V = V.numel() > 0;
-
     auto stream = c10::hip::getCurrentHIPStream().stream();
     hipError_t err = skdist_sgd_step(
         Xs.data_ptr(), XsT.data_ptr(), WbfT.data_ptr(), GT.data_ptr(),
         W.data_ptr(), has_V ? V.data_ptr() : nullptr, WbfT.data_ptr(),
         partial.data_ptr(), y.data_ptr(), fold.data_ptr(),
         col_class.data_ptr(), col_fold.data_ptr(), col_lr.data_ptr(),
-        col_l2.data_ptr(), start, m, n, (int)fa, (int)ncols_pad,
-        (int)gt_stride, (int)splitk, (int)loss_id, (float)lr_scale,
-        (float)momentum, (int)intercept_row, stream);
+        col_l2.data_ptr(), start, m, a.n, a.n_pad, a.fa_store, (int)a.fa,
+        (int)a.ncols_pad, (int)a.gt_stride, (int)a.splitk, (int)loss_id,
+        (float)lr_scale, (float)momentum, (int)intercept_row, stream);
     TORCH_CHECK(err == hipSuccess, "skdist_sgd_step: ",
                 hipGetErrorString(err));
 }
@@ -68,23 +83,21 @@ void sgd_epoch(torch::Tensor Xs, torch::Tensor XsT, torch::Tensor GT,
                torch::Tensor col_lr, torch::Tensor col_l2,
                int64_t batch_size, int64_t loss_id, double lr_scale,
                double momentum, int64_t intercept_row) {
-    const auto n = Xs.size(0);
-    const auto fa = Xs.size(1);
-    const auto ncols_pad = W.size(1);
-    const auto gt_stride = GT.size(1);
-    const auto splitk = partial.size(0);
+    auto a = check_args(Xs, XsT, GT, W, partial, y);
+    TORCH_CHECK(batch_size % 128 == 0, "batch_size must be a mult of 128");
     const bool has_V = V.numel() > 0;
     auto stream = c10::hip::getCurrentHIPStream().stream();
-    for (int64_t start = 0; start < n; start += batch_size) {
-        const int64_t m = std::min(batch_size, n - start);
+    for (int64_t start = 0; start < a.n; start += batch_size) {
+        const int64_t m = std::min(batch_size, a.n - start);
         hipError_t err = skdist_sgd_step(
             Xs.data_ptr(), XsT.data_ptr(), WbfT.data_ptr(), GT.data_ptr(),
             W.data_ptr(), has_V ? V.data_ptr() : nullptr, WbfT.data_ptr(),
             partial.data_ptr(), y.data_ptr(), fold.data_ptr(),
             col_class.data_ptr(), col_fold.data_ptr(), col_lr.data_ptr(),
-            col_l2.data_ptr(), start, m, n, (int)fa, (int)ncols_pad,
-            (int)gt_stride, (int)splitk, (int)loss_id, (float)lr_scale,
-            (float)momentum, (int)intercept_row, stream);
+            col_l2.data_ptr(), start, m, a.n, a.n_pad, a.fa_store,
+            (int)a.fa, (int)a.ncols_pad, (int)a.gt_stride, (int)a.splitk,
+            (int)loss_id, (float)lr_scale, (float)momentum,
+            (int)intercept_row, stream);
         TORCH_CHECK(err == hipSuccess, "sgd_epoch: ", hipGetErrorString(err));
     }
 }
