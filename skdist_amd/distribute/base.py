@@ -53,12 +53,13 @@ def _parse_partitions(partitions, n_tasks):
 
 
 def _strip_sc(est):
-    """Remove scheduler handles before returning a fitted estimator
-    (pickle contract, reference search.py:568-570, multiclass.py:283-285).
+    """Null scheduler handles before returning a fitted estimator
+    (pickle contract, reference search.py:568-570, multiclass.py:283-285 —
+    we keep the attribute as None so get_params/clone still work).
     """
-    if hasattr(est, "sc"):
-        del est.sc
+    if getattr(est, "sc", None) is not None:
+        est.sc = None
     nested = getattr(est, "estimator", None)
-    if nested is not None and hasattr(nested, "sc"):
-        del nested.sc
+    if nested is not None and getattr(nested, "sc", None) is not None:
+        nested.sc = None
     return est

@@ -204,9 +204,14 @@ class DeviceDataset:
 
 
 class ColumnSpec:
-    """Per-column metadata for one batched solve (device tensors)."""
+    """Per-column metadata for one batched solve (device tensors).
 
-    def __init__(self, device, col_fold, col_class, col_lr, col_l2):
+    ``col_class2[c] >= 0`` marks a one-vs-one column: only rows with
+    ``y in (col_class[c], col_class2[c])`` train it.
+    """
+
+    def __init__(self, device, col_fold, col_class, col_lr, col_l2,
+                 col_class2=None):
         as_t = lambda a, dt: torch.as_tensor(
             np.ascontiguousarray(a), dtype=dt, device=device
         )
@@ -214,6 +219,9 @@ class ColumnSpec:
         self.col_class = as_t(col_class, torch.int32)
         self.col_lr = as_t(col_lr, torch.float32)
         self.col_l2 = as_t(col_l2, torch.float32)
+        if col_class2 is None:
+            col_class2 = np.full(len(col_fold), -1, dtype=np.int32)
+        self.col_class2 = as_t(col_class2, torch.int32)
         self.ncols = len(col_fold)
 
 
@@ -278,10 +286,14 @@ def _sgd_step_torch(Xaug, y_float, fold_id, idx, W, V, spec, loss_id,
         G = Z - t
 
     if fold_id is not None:
-        mask = (
-            fold_id[idx].unsqueeze(1) != spec.col_fold.unsqueeze(0)
-        ).to(torch.float32)
-        G = G * mask
+        mask = fold_id[idx].unsqueeze(1) != spec.col_fold.unsqueeze(0)
+        c2 = spec.col_class2.unsqueeze(0)
+        pair_ok = (
+            (c2 < 0)
+            | (yb.unsqueeze(1) == spec.col_class.unsqueeze(0).float())
+            | (yb.unsqueeze(1) == c2.float())
+        )
+        G = G * (mask & pair_ok).to(torch.float32)
 
     grad = (Xb.transpose(0, 1).to(comp) @ G.to(comp)).to(torch.float32) / m
     # L2 on weights only, not the intercept row

@@ -13,6 +13,7 @@ contract, reference search.py:568-570).
 """
 
 import time
+from itertools import combinations
 
 import numpy as np
 from sklearn.base import BaseEstimator, ClassifierMixin, RegressorMixin
@@ -271,6 +272,69 @@ class _BatchedLinearBase(BaseEstimator):
             out = [results[i] for i in range(n_cand * n_folds)]
             refit_all = refit_local
         return {"tasks": out, "refit_estimators": refit_all}
+
+    def batched_multiclass_fit(self, X, y, cluster, mode="ovr"):
+        """Train every one-vs-rest class (or one-vs-one pair) binary
+        problem as one batched device solve; used by
+        DistOneVsRest/OneVsOneClassifier.  Returns (classes, estimators)
+        with estimators[i] a fitted BINARY copy of self (classes_=[0,1],
+        positive class = the OvR class / the pair's second class).
+        """
+        t0 = time.perf_counter()
+        ds = DeviceDataset(
+            None if X is None else np.asarray(X, dtype=np.float32),
+            None if y is None else np.asarray(y),
+            cluster=cluster, standardize=self.standardize,
+        )
+        ds.set_cv_partition([])
+        classes = ds.classes_
+        if classes is None:
+            raise FallbackToGeneric("targets look continuous")
+        k = len(classes)
+        if mode == "ovr":
+            problems = [(i, -1) for i in range(k)]
+        else:
+            problems = [(j, i) for i, j in combinations(range(k), 2)]
+        ids = (
+            cluster.shard_indices(len(problems)) if cluster is not None
+            else list(range(len(problems)))
+        )
+        counts = np.bincount(ds.y_int.cpu().numpy(), minlength=k)
+        col_class, col_class2, col_lr, col_l2 = [], [], [], []
+        for pi in ids:
+            tgt, other = problems[pi]
+            n_train = ds.n if other < 0 else counts[tgt] + counts[other]
+            col_class.append(tgt)
+            col_class2.append(other)
+            col_lr.append(self.lr)
+            col_l2.append(self._lam(n_train))
+        local = {}
+        if ids:
+            spec = ColumnSpec(
+                ds.device,
+                col_fold=np.full(len(ids), -2, dtype=np.int32),
+                col_class=np.asarray(col_class, dtype=np.int32),
+                col_lr=np.asarray(col_lr, dtype=np.float32),
+                col_l2=np.asarray(col_l2, dtype=np.float32),
+                col_class2=np.asarray(col_class2, dtype=np.int32),
+            )
+            W = batched_sgd_fit(
+                ds, spec, self._loss, self.epochs, self.batch_size,
+                seed=self._seed(), momentum=self.momentum,
+            )
+            per = (time.perf_counter() - t0) / len(ids)
+            for ci, pi in enumerate(ids):
+                est = sk_clone_without_sc(self)
+                est._store_fitted(ds, W[:, ci : ci + 1], True)
+                est.classes_ = np.array([0, 1])
+                est.n_features_in_ = ds.f
+                est.fit_time_ = per
+                local[pi] = est
+        if cluster is not None:
+            ests = cluster.gather_task_results(local, len(problems))
+        else:
+            ests = [local[i] for i in range(len(problems))]
+        return classes, ests
 
     def _device_metric(self, scoring):
         if scoring is None:

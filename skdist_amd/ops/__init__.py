@@ -85,6 +85,7 @@ def hip_sgd_solve(ds, spec, loss_id, epochs, batch_size, seed, momentum,
                           device=device)
     cls_p = _pad_cols(spec.col_class, ncp, -99)
     cfold_p = _pad_cols(spec.col_fold, ncp, -9)
+    cls2_p = _pad_cols(spec.col_class2, ncp, -1)
     lr_p = _pad_cols(spec.col_lr, ncp, 0.0)
     l2_p = _pad_cols(spec.col_l2, ncp, 0.0)
 
@@ -99,7 +100,7 @@ def hip_sgd_solve(ds, spec, loss_id, epochs, batch_size, seed, momentum,
         lr_scale = 1.0 / (1.0 + lr_decay * epoch)
         ext.sgd_epoch(
             Xs, XsT, GT, W, V, WbfT, partial, ys, folds,
-            cls_p, cfold_p, lr_p, l2_p,
+            cls_p, cfold_p, cls2_p, lr_p, l2_p,
             bs, int(loss_id), float(lr_scale), float(momentum),
             int(ds.intercept_row),
         )

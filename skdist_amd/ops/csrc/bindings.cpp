@@ -9,7 +9,7 @@ extern "C" hipError_t skdist_sgd_step(
     const void* Xs, const void* XsT, const void* WbfT_in,
     void* GT, void* W, void* V, void* WbfT, void* partial,
     const void* y, const void* fold,
-    const void* col_class, const void* col_fold,
+    const void* col_class, const void* col_fold, const void* col_class2,
     const void* col_lr, const void* col_l2,
     long long start, long long m, long long n, long long n_pad,
     long long fa_store, int fa, int ncols_pad,
@@ -51,6 +51,7 @@ void sgd_step(torch::Tensor Xs, torch::Tensor XsT, torch::Tensor GT,
               torch::Tensor W, torch::Tensor V, torch::Tensor WbfT,
               torch::Tensor partial, torch::Tensor y, torch::Tensor fold,
               torch::Tensor col_class, torch::Tensor col_fold,
+              torch::Tensor col_class2,
               torch::Tensor col_lr, torch::Tensor col_l2,
               int64_t start, int64_t m, int64_t loss_id, double lr_scale,
               double momentum, int64_t intercept_row) {
@@ -66,7 +67,8 @@ void sgd_step(torch::Tensor Xs, torch::Tensor XsT, torch::Tensor GT,
         Xs.data_ptr(), XsT.data_ptr(), WbfT.data_ptr(), GT.data_ptr(),
         W.data_ptr(), has_V ? V.data_ptr() : nullptr, WbfT.data_ptr(),
         partial.data_ptr(), y.data_ptr(), fold.data_ptr(),
-        col_class.data_ptr(), col_fold.data_ptr(), col_lr.data_ptr(),
+        col_class.data_ptr(), col_fold.data_ptr(), col_class2.data_ptr(),
+        col_lr.data_ptr(),
         col_l2.data_ptr(), start, m, a.n, a.n_pad, a.fa_store, (int)a.fa,
         (int)a.ncols_pad, (int)a.gt_stride, (int)a.splitk, (int)loss_id,
         (float)lr_scale, (float)momentum, (int)intercept_row, stream);
@@ -80,6 +82,7 @@ void sgd_epoch(torch::Tensor Xs, torch::Tensor XsT, torch::Tensor GT,
                torch::Tensor W, torch::Tensor V, torch::Tensor WbfT,
                torch::Tensor partial, torch::Tensor y, torch::Tensor fold,
                torch::Tensor col_class, torch::Tensor col_fold,
+               torch::Tensor col_class2,
                torch::Tensor col_lr, torch::Tensor col_l2,
                int64_t batch_size, int64_t loss_id, double lr_scale,
                double momentum, int64_t intercept_row) {
@@ -93,7 +96,8 @@ void sgd_epoch(torch::Tensor Xs, torch::Tensor XsT, torch::Tensor GT,
             Xs.data_ptr(), XsT.data_ptr(), WbfT.data_ptr(), GT.data_ptr(),
             W.data_ptr(), has_V ? V.data_ptr() : nullptr, WbfT.data_ptr(),
             partial.data_ptr(), y.data_ptr(), fold.data_ptr(),
-            col_class.data_ptr(), col_fold.data_ptr(), col_lr.data_ptr(),
+            col_class.data_ptr(), col_fold.data_ptr(),
+            col_class2.data_ptr(), col_lr.data_ptr(),
             col_l2.data_ptr(), start, m, a.n, a.n_pad, a.fa_store,
             (int)a.fa, (int)a.ncols_pad, (int)a.gt_stride, (int)a.splitk,
             (int)loss_id, (float)lr_scale, (float)momentum,

@@ -66,6 +66,7 @@ extern "C" __global__ __launch_bounds__(256) void k_fwd_gt(
     const int* __restrict__ fold,     // [n]
     const int* __restrict__ col_class,
     const int* __restrict__ col_fold,
+    const int* __restrict__ col_class2,  // >=0: one-vs-one partner class
     int start, int m, long long n, int fa, int ncols_pad, int gt_stride,
     int loss_id)
 {
@@ -78,6 +79,7 @@ extern "C" __global__ __launch_bounds__(256) void k_fwd_gt(
     int* fold_s = (int*)(meta + 512);       // [BM]
     int* cls_s = (int*)(meta + 1024);       // [BN]
     int* cfold_s = (int*)(meta + 1536);     // [BN]
+    int* cls2_s = (int*)(meta + 2048);      // [BN]
 
     const int tid = threadIdx.x;
     const int bm = blockIdx.x * BM;
@@ -94,6 +96,7 @@ extern "C" __global__ __launch_bounds__(256) void k_fwd_gt(
         const int c = tid - BM;
         cls_s[c] = col_class[bn + c];
         cfold_s[c] = col_fold[bn + c];
+        cls2_s[c] = col_class2[bn + c];
     }
 
     const __bf16* Abase = Xs + (long long)(start + bm) * fa;
@@ -144,6 +147,7 @@ extern "C" __global__ __launch_bounds__(256) void k_fwd_gt(
             const int colb = wc + ni * 16 + fr;
             const int cls = cls_s[colb];
             const int cfo = cfold_s[colb];
+            const int c2 = cls2_s[colb];
             short4v g4;
             #pragma unroll
             for (int r = 0; r < 4; ++r) {
@@ -153,7 +157,10 @@ extern "C" __global__ __launch_bounds__(256) void k_fwd_gt(
                 const float t =
                     (cls < 0) ? yv : (yv == (float)cls ? 1.f : 0.f);
                 const float g = dloss(loss_id, z, t);
-                const bool train = (fold_s[row] != cfo) && (bm + row < m);
+                bool train = (fold_s[row] != cfo) && (bm + row < m);
+                if (c2 >= 0)  // one-vs-one: only the pair's rows train
+                    train = train &&
+                            (yv == (float)cls || yv == (float)c2);
                 g4[r] = __builtin_bit_cast(short,
                                            f32_to_bf16(train ? g : 0.f));
             }
@@ -296,7 +303,7 @@ extern "C" hipError_t skdist_sgd_step(
     const void* Xs, const void* XsT, const void* WbfT_in,
     void* GT, void* W, void* V, void* WbfT, void* partial,
     const void* y, const void* fold,
-    const void* col_class, const void* col_fold,
+    const void* col_class, const void* col_fold, const void* col_class2,
     const void* col_lr, const void* col_l2,
     long long start, long long m, long long n, long long n_pad,
     long long fa_store, int fa, int ncols_pad,
@@ -307,11 +314,12 @@ extern "C" hipError_t skdist_sgd_step(
     const int m_pad = (int)((m + BM - 1) / BM) * BM;
     {
         dim3 grid(m_pad / BM, ncols_pad / BN);
-        size_t lds = (size_t)BN * LDC * 2 + 2048;
+        size_t lds = (size_t)BN * LDC * 2 + 2560;
         hipLaunchKernelGGL(k_fwd_gt, grid, dim3(256), lds, stream,
                            (const __bf16*)Xs, (const __bf16*)WbfT_in,
                            (__bf16*)GT, (const float*)y, (const int*)fold,
                            (const int*)col_class, (const int*)col_fold,
+                           (const int*)col_class2,
                            (int)start, (int)m, n, fa, ncols_pad, gt_stride,
                            loss_id);
         HIP_CHECK(hipGetLastError());

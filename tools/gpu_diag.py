@@ -44,10 +44,13 @@ def main():
     splitk = 4
     partial = torch.empty(splitk, fa, ncols, device=dev)
 
-    XsT = X.t().contiguous()
+    fa_store = (fa + 127) // 128 * 128
+    XsT = torch.zeros(fa_store, n, dtype=torch.bfloat16, device=dev)
+    XsT[:fa] = X.t()
     W0 = W.clone()
+    col_class2 = torch.full((ncols,), -1, dtype=torch.int32, device=dev)
     ext.sgd_step(X, XsT, GT, W, V, WbfT, partial, y, fold,
-                 col_class, col_fold, col_lr, col_l2,
+                 col_class, col_fold, col_class2, col_lr, col_l2,
                  0, n, 2, 1.0, 0.0, fa - 1)
     torch.cuda.synchronize()
 
