@@ -1,0 +1,395 @@
+"""
+Distributed tree ensembles (reference: skdist/distribute/ensemble.py).
+
+One tree per task (reference parallel axis P5): the
+scheduler shards per-tree seeds round-robin across ranks; each task fits
+one decision tree against the rank-local copy of (X, y) with bootstrap
+weights, and the pickled trees gather back (tiny vs the data, which never
+moves after the one-time broadcast).
+
+Unlike the reference — which re-implemented sklearn's private forest
+internals and pinned sklearn <0.23.2 (ensemble.py:11-19) — the trees here
+are plain public ``DecisionTree*``/``ExtraTree*`` estimators and the
+forest aggregation (proba mean / prediction mean / leaf one-hot) is our
+own, so any modern sklearn works.  OOB scoring is implemented for real
+(the reference's ``_set_oob_score`` is deliberately a no-op,
+ensemble.py:338-340).
+
+The batched HIP histogram tree builder (skdist_amd.models.forest) plugs
+into the same classes as a device fast path.
+"""
+
+import numpy as np
+import scipy.sparse as sp
+from sklearn.base import BaseEstimator, ClassifierMixin, RegressorMixin, TransformerMixin
+from sklearn.preprocessing import OneHotEncoder
+from sklearn.tree import (
+    DecisionTreeClassifier,
+    DecisionTreeRegressor,
+    ExtraTreeClassifier,
+    ExtraTreeRegressor,
+)
+from sklearn.utils.validation import check_random_state
+
+from ..parallel.local import run_local_tasks
+from .base import _strip_sc
+from .validation import _check_estimator
+
+MAX_SEED = np.iinfo(np.int32).max
+
+
+def _bootstrap_weights(seed, n, base_weight=None):
+    """Multinomial bootstrap as sample weights (reference
+    ensemble.py:51-55,95-97 semantics: randint + bincount)."""
+    rng = np.random.RandomState(seed)
+    counts = np.bincount(rng.randint(0, n, n), minlength=n).astype(np.float64)
+    if base_weight is not None:
+        counts *= base_weight
+    return counts
+
+
+def _resolve_max_features(max_features, is_classifier):
+    # the reference era's 'auto' (sqrt for clf, n for reg); modern sklearn
+    # rejects 'auto'
+    if max_features == "auto":
+        return "sqrt" if is_classifier else 1.0
+    return max_features
+
+
+def _fit_one_tree(tree_proto, X, y, seed, bootstrap, sample_weight,
+                  class_weight=None):
+    """Fit one tree task (reference worker _build_trees,
+    ensemble.py:68-109)."""
+    tree = tree_proto.__class__(**tree_proto.get_params())
+    tree.set_params(random_state=seed)
+    if bootstrap:
+        w = _bootstrap_weights(seed, X.shape[0], sample_weight)
+    else:
+        w = sample_weight
+    tree.fit(X, y, sample_weight=w)
+    return tree
+
+
+class DistBaseForest(BaseEstimator):
+    """Shared fan-out machinery for all forest classes
+    (reference ensemble.py:154-340)."""
+
+    _is_classifier = False
+
+    def _tree_proto(self):
+        raise NotImplementedError
+
+    def fit(self, X, y, sample_weight=None):
+        _check_estimator(self, verbose=bool(self.verbose))
+        sc = self.sc
+        if sc is not None and getattr(sc, "distributed", False):
+            X, y, sample_weight = sc.sync_host_data(X, y, sample_weight)
+        X = np.asarray(X) if not sp.issparse(X) else X.tocsr()
+        y = np.asarray(y)
+        if self._is_classifier:
+            self.classes_ = np.unique(y)
+            self.n_classes_ = len(self.classes_)
+        self.n_features_in_ = X.shape[1]
+
+        rnd = check_random_state(self.random_state)
+        seeds = rnd.randint(MAX_SEED, size=self.n_estimators)
+        proto = self._tree_proto()
+
+        if not self.warm_start or not hasattr(self, "estimators_"):
+            self.estimators_ = []
+        n_more = self.n_estimators - len(self.estimators_)
+        if n_more < 0:
+            raise ValueError(
+                f"n_estimators={self.n_estimators} must be >= "
+                f"len(estimators_)={len(self.estimators_)} when warm_start"
+            )
+        seeds = seeds[:n_more]
+
+        def task_fn(task):
+            i, seed = task
+            return i, _fit_one_tree(
+                proto, X, y, int(seed), self.bootstrap, sample_weight
+            )
+
+        tasks = list(enumerate(seeds))
+        if sc is None:
+            results = run_local_tasks(task_fn, tasks, n_jobs=self.n_jobs)
+        else:
+            results = sc.run_tasks(task_fn, tasks)
+        results.sort(key=lambda t: t[0])
+        self.estimators_.extend(r[1] for r in results)
+        self._seeds = [int(s) for s in seeds]
+
+        if self.oob_score:
+            self._compute_oob(X, y)
+        _strip_sc(self)
+        return self
+
+    # ------------------------------------------------------------------ #
+    def _compute_oob(self, X, y):
+        """Real OOB scoring (the reference stubs this out,
+        ensemble.py:338-340)."""
+        n = X.shape[0]
+        if self._is_classifier:
+            agg = np.zeros((n, self.n_classes_))
+        else:
+            agg = np.zeros(n)
+        cnt = np.zeros(n)
+        for seed, tree in zip(self._seeds, self.estimators_):
+            counts = _bootstrap_weights(seed, n) if self.bootstrap else None
+            oob = (
+                np.flatnonzero(counts == 0) if counts is not None
+                else np.arange(n)
+            )
+            if len(oob) == 0:
+                continue
+            if self._is_classifier:
+                agg[oob] += tree.predict_proba(X[oob])
+            else:
+                agg[oob] += tree.predict(X[oob])
+            cnt[oob] += 1
+        seen = cnt > 0
+        if self._is_classifier:
+            pred = self.classes_[agg[seen].argmax(axis=1)]
+            self.oob_score_ = float(np.mean(pred == y[seen]))
+            with np.errstate(invalid="ignore"):
+                self.oob_decision_function_ = agg / np.maximum(
+                    cnt[:, None], 1
+                )
+        else:
+            pred = agg[seen] / cnt[seen]
+            ss_res = float(np.sum((y[seen] - pred) ** 2))
+            ss_tot = float(np.sum((y[seen] - y[seen].mean()) ** 2))
+            self.oob_score_ = 1.0 - ss_res / max(ss_tot, 1e-300)
+            self.oob_prediction_ = np.where(
+                seen, agg / np.maximum(cnt, 1), np.nan
+            )
+
+    # ------------------------------------------------------------------ #
+    def apply(self, X):
+        return np.column_stack([t.apply(X) for t in self.estimators_])
+
+    @property
+    def feature_importances_(self):
+        imp = np.mean(
+            [t.feature_importances_ for t in self.estimators_], axis=0
+        )
+        s = imp.sum()
+        return imp / s if s > 0 else imp
+
+
+class _ForestClassifierMixin(ClassifierMixin):
+    _is_classifier = True
+
+    def predict_proba(self, X):
+        proba = None
+        for tree in self.estimators_:
+            p = tree.predict_proba(X)
+            # align tree classes to forest classes (trees see full y via
+            # bootstrap sample weights, so classes always match)
+            proba = p if proba is None else proba + p
+        return proba / len(self.estimators_)
+
+    def predict(self, X):
+        return self.classes_[self.predict_proba(X).argmax(axis=1)]
+
+    def predict_log_proba(self, X):
+        return np.log(np.clip(self.predict_proba(X), 1e-300, None))
+
+
+class _ForestRegressorMixin(RegressorMixin):
+    _is_classifier = False
+
+    def predict(self, X):
+        out = None
+        for tree in self.estimators_:
+            p = tree.predict(X)
+            out = p if out is None else out + p
+        return out / len(self.estimators_)
+
+
+_COMMON_TREE_PARAMS = (
+    "criterion", "max_depth", "min_samples_split", "min_samples_leaf",
+    "min_weight_fraction_leaf", "max_leaf_nodes", "min_impurity_decrease",
+)
+
+
+def _forest_init(self, sc, partitions, n_estimators, bootstrap, oob_score,
+                 n_jobs, random_state, verbose, warm_start, class_weight,
+                 tree_kwargs):
+    self.sc = sc
+    self.partitions = partitions
+    self.n_estimators = n_estimators
+    self.bootstrap = bootstrap
+    self.oob_score = oob_score
+    self.n_jobs = n_jobs
+    self.random_state = random_state
+    self.verbose = verbose
+    self.warm_start = warm_start
+    self.class_weight = class_weight
+    for k, v in tree_kwargs.items():
+        setattr(self, k, v)
+
+
+class DistRandomForestClassifier(_ForestClassifierMixin, DistBaseForest):
+    """Distributed random forest classifier (reference ensemble.py:365-421)."""
+
+    def __init__(self, sc=None, partitions="auto", n_estimators=100,
+                 criterion="gini", max_depth=None, min_samples_split=2,
+                 min_samples_leaf=1, min_weight_fraction_leaf=0.0,
+                 max_features="auto", max_leaf_nodes=None,
+                 min_impurity_decrease=0.0, bootstrap=True, oob_score=False,
+                 n_jobs=None, random_state=None, verbose=0, warm_start=False,
+                 class_weight=None):
+        _forest_init(self, sc, partitions, n_estimators, bootstrap,
+                     oob_score, n_jobs, random_state, verbose, warm_start,
+                     class_weight, dict(
+                         criterion=criterion, max_depth=max_depth,
+                         min_samples_split=min_samples_split,
+                         min_samples_leaf=min_samples_leaf,
+                         min_weight_fraction_leaf=min_weight_fraction_leaf,
+                         max_features=max_features,
+                         max_leaf_nodes=max_leaf_nodes,
+                         min_impurity_decrease=min_impurity_decrease))
+
+    def _tree_proto(self):
+        return DecisionTreeClassifier(
+            **{k: getattr(self, k) for k in _COMMON_TREE_PARAMS},
+            class_weight=self.class_weight,
+            max_features=_resolve_max_features(self.max_features, True),
+        )
+
+
+class DistExtraTreesClassifier(_ForestClassifierMixin, DistBaseForest):
+    """Distributed extra-trees classifier (reference ensemble.py:424-480)."""
+
+    def __init__(self, sc=None, partitions="auto", n_estimators=100,
+                 criterion="gini", max_depth=None, min_samples_split=2,
+                 min_samples_leaf=1, min_weight_fraction_leaf=0.0,
+                 max_features="auto", max_leaf_nodes=None,
+                 min_impurity_decrease=0.0, bootstrap=False, oob_score=False,
+                 n_jobs=None, random_state=None, verbose=0, warm_start=False,
+                 class_weight=None):
+        _forest_init(self, sc, partitions, n_estimators, bootstrap,
+                     oob_score, n_jobs, random_state, verbose, warm_start,
+                     class_weight, dict(
+                         criterion=criterion, max_depth=max_depth,
+                         min_samples_split=min_samples_split,
+                         min_samples_leaf=min_samples_leaf,
+                         min_weight_fraction_leaf=min_weight_fraction_leaf,
+                         max_features=max_features,
+                         max_leaf_nodes=max_leaf_nodes,
+                         min_impurity_decrease=min_impurity_decrease))
+
+    def _tree_proto(self):
+        return ExtraTreeClassifier(
+            **{k: getattr(self, k) for k in _COMMON_TREE_PARAMS},
+            class_weight=self.class_weight,
+            max_features=_resolve_max_features(self.max_features, True),
+        )
+
+
+class DistRandomForestRegressor(_ForestRegressorMixin, DistBaseForest):
+    """Distributed random forest regressor (reference ensemble.py:505-559)."""
+
+    def __init__(self, sc=None, partitions="auto", n_estimators=100,
+                 criterion="squared_error", max_depth=None,
+                 min_samples_split=2, min_samples_leaf=1,
+                 min_weight_fraction_leaf=0.0, max_features="auto",
+                 max_leaf_nodes=None, min_impurity_decrease=0.0,
+                 bootstrap=True, oob_score=False, n_jobs=None,
+                 random_state=None, verbose=0, warm_start=False):
+        _forest_init(self, sc, partitions, n_estimators, bootstrap,
+                     oob_score, n_jobs, random_state, verbose, warm_start,
+                     None, dict(
+                         criterion=criterion, max_depth=max_depth,
+                         min_samples_split=min_samples_split,
+                         min_samples_leaf=min_samples_leaf,
+                         min_weight_fraction_leaf=min_weight_fraction_leaf,
+                         max_features=max_features,
+                         max_leaf_nodes=max_leaf_nodes,
+                         min_impurity_decrease=min_impurity_decrease))
+
+    def _tree_proto(self):
+        return DecisionTreeRegressor(
+            **{k: getattr(self, k) for k in _COMMON_TREE_PARAMS},
+            max_features=_resolve_max_features(self.max_features, False),
+        )
+
+
+class DistExtraTreesRegressor(_ForestRegressorMixin, DistBaseForest):
+    """Distributed extra-trees regressor (reference ensemble.py:562-616)."""
+
+    def __init__(self, sc=None, partitions="auto", n_estimators=100,
+                 criterion="squared_error", max_depth=None,
+                 min_samples_split=2, min_samples_leaf=1,
+                 min_weight_fraction_leaf=0.0, max_features="auto",
+                 max_leaf_nodes=None, min_impurity_decrease=0.0,
+                 bootstrap=False, oob_score=False, n_jobs=None,
+                 random_state=None, verbose=0, warm_start=False):
+        _forest_init(self, sc, partitions, n_estimators, bootstrap,
+                     oob_score, n_jobs, random_state, verbose, warm_start,
+                     None, dict(
+                         criterion=criterion, max_depth=max_depth,
+                         min_samples_split=min_samples_split,
+                         min_samples_leaf=min_samples_leaf,
+                         min_weight_fraction_leaf=min_weight_fraction_leaf,
+                         max_features=max_features,
+                         max_leaf_nodes=max_leaf_nodes,
+                         min_impurity_decrease=min_impurity_decrease))
+
+    def _tree_proto(self):
+        return ExtraTreeRegressor(
+            **{k: getattr(self, k) for k in _COMMON_TREE_PARAMS},
+            max_features=_resolve_max_features(self.max_features, False),
+        )
+
+
+class DistRandomTreesEmbedding(TransformerMixin, DistBaseForest):
+    """Distributed totally-random-trees embedding
+    (reference ensemble.py:619-717)."""
+
+    _is_classifier = False
+
+    def __init__(self, sc=None, partitions="auto", n_estimators=100,
+                 max_depth=5, min_samples_split=2, min_samples_leaf=1,
+                 min_weight_fraction_leaf=0.0, max_leaf_nodes=None,
+                 min_impurity_decrease=0.0, sparse_output=True, n_jobs=None,
+                 random_state=None, verbose=0, warm_start=False):
+        _forest_init(self, sc, partitions, n_estimators, False, False,
+                     n_jobs, random_state, verbose, warm_start, None, dict(
+                         max_depth=max_depth,
+                         min_samples_split=min_samples_split,
+                         min_samples_leaf=min_samples_leaf,
+                         min_weight_fraction_leaf=min_weight_fraction_leaf,
+                         max_leaf_nodes=max_leaf_nodes,
+                         min_impurity_decrease=min_impurity_decrease,
+                         sparse_output=sparse_output))
+
+    def _tree_proto(self):
+        return ExtraTreeRegressor(
+            criterion="squared_error", max_depth=self.max_depth,
+            min_samples_split=self.min_samples_split,
+            min_samples_leaf=self.min_samples_leaf,
+            min_weight_fraction_leaf=self.min_weight_fraction_leaf,
+            max_features=1, max_leaf_nodes=self.max_leaf_nodes,
+            min_impurity_decrease=self.min_impurity_decrease,
+        )
+
+    def fit(self, X, y=None, sample_weight=None):
+        self.fit_transform(X, y, sample_weight=sample_weight)
+        return self
+
+    def fit_transform(self, X, y=None, sample_weight=None):
+        X = np.asarray(X) if not sp.issparse(X) else X.tocsc()
+        rnd = check_random_state(self.random_state)
+        y_rand = rnd.uniform(size=X.shape[0])
+        DistBaseForest.fit(self, X, y_rand, sample_weight=sample_weight)
+        self.one_hot_encoder_ = OneHotEncoder(
+            sparse_output=self.sparse_output, categories="auto",
+            handle_unknown="ignore",
+        )
+        return self.one_hot_encoder_.fit_transform(self.apply(X))
+
+    def transform(self, X):
+        return self.one_hot_encoder_.transform(self.apply(X))
