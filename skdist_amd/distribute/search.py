@@ -416,6 +416,181 @@ class DistGridSearchCV(DistBaseSearchCV):
         return ParameterGrid(self.param_grid)
 
 
+class DistMultiModelSearch(BaseEstimator, MetaEstimatorMixin):
+    """Randomized search across heterogeneous model families in one task
+    pool (reference search.py:717-908).
+
+    ``models`` is a list of ``(name, estimator, param_distributions[, n])``
+    tuples; ``n`` (or the global ``n``) parameter sets are sampled per
+    model and every (model × params × fold) combination is one task.
+
+    Reference bugs fixed here: per-model seeded sampling
+    (the reference's ``_sample_generator`` referenced an undefined loop
+    variable, search.py:809-811) and ``worst_score_`` actually reporting
+    the worst score (search.py:836-837 set both from the best row).
+    """
+
+    def __init__(self, models, sc=None, partitions="auto", n=5, cv=5,
+                 scoring=None, random_state=None, verbose=0, refit=True,
+                 n_jobs=None, pre_dispatch="2*n_jobs"):
+        self.models = models
+        self.sc = sc
+        self.partitions = partitions
+        self.n = n
+        self.cv = cv
+        self.scoring = scoring
+        self.random_state = random_state
+        self.verbose = verbose
+        self.refit = refit
+        self.n_jobs = n_jobs
+        self.pre_dispatch = pre_dispatch
+
+    def fit(self, X, y=None, groups=None, **fit_params):
+        from .validation import _check_n_iter, _validate_models
+
+        _check_estimator(self, verbose=self.verbose)
+        sc = self.sc
+        if sc is not None and getattr(sc, "distributed", False):
+            X, y, groups = sc.sync_host_data(X, y, groups)
+        models = _validate_models(self.models, self)
+        cv = check_cv(self.cv, y, classifier=is_classifier(models[0][1]))
+        X, y, groups = indexable(X, y, groups)
+        folds = list(cv.split(X, y, groups))
+
+        # sample parameter sets per model (deterministic per-model seeds)
+        entries = []  # (model_idx, params_idx, params)
+        for mi, (name, est, dists, n_override) in enumerate(models):
+            n_iter = _check_n_iter(
+                n_override if n_override is not None else self.n, dists
+            )
+            seed = (
+                None if self.random_state is None
+                else self.random_state + mi
+            )
+            for pi, params in enumerate(
+                ParameterSampler(dists, n_iter, random_state=seed)
+            ):
+                entries.append((mi, pi, params))
+
+        tasks = []
+        tid = 0
+        for mi, pi, params in entries:
+            for fi, split in enumerate(folds):
+                tasks.append((tid, mi, pi, params, split))
+                tid += 1
+
+        def task_fn(task):
+            tid_, mi, pi, params, (train, test) = task
+            est = _clone(models[mi][1])
+            if hasattr(est, "sc"):
+                est.sc = None
+            if params:
+                est.set_params(**params)
+            X_tr, y_tr = _safe_split(est, X, y, train)
+            X_te, y_te = _safe_split(est, X, y, test, train)
+            est.fit(X_tr, y_tr, **fit_params)
+            from sklearn.metrics import check_scoring
+
+            scorer = check_scoring(est, scoring=self.scoring)
+            return tid_, mi, pi, float(scorer(est, X_te, y_te))
+
+        if sc is None:
+            results = run_local_tasks(
+                task_fn, tasks, n_jobs=self.n_jobs,
+                pre_dispatch=self.pre_dispatch,
+            )
+        else:
+            results = sc.run_tasks(task_fn, tasks)
+
+        # aggregate mean score per (model, params)
+        agg = {}
+        for tid_, mi, pi, score in results:
+            agg.setdefault((mi, pi), []).append(score)
+        rows = []
+        for (mi, pi, params) in entries:
+            scores = agg[(mi, pi)]
+            rows.append({
+                "model_index": mi,
+                "model_name": models[mi][0],
+                "params": params,
+                "mean_test_score": float(np.mean(scores)),
+                "std_test_score": float(np.std(scores)),
+            })
+        means = np.array([r["mean_test_score"] for r in rows])
+        from scipy.stats import rankdata
+
+        ranks = np.asarray(rankdata(-means, method="min"), dtype=np.int32)
+        self.cv_results_ = {
+            "model_index": [r["model_index"] for r in rows],
+            "model_name": [r["model_name"] for r in rows],
+            "params": [r["params"] for r in rows],
+            "rank_test_score": ranks,
+            "mean_test_score": means,
+            "std_test_score": np.array(
+                [r["std_test_score"] for r in rows]
+            ),
+        }
+        best = int(means.argmax())
+        self.best_index_ = best
+        self.best_model_index_ = rows[best]["model_index"]
+        self.best_model_name_ = rows[best]["model_name"]
+        self.best_params_ = rows[best]["params"]
+        self.best_score_ = float(means[best])
+        self.worst_score_ = float(means.min())
+
+        if self.verbose:
+            per_model = {}
+            for r in rows:
+                per_model[r["model_name"]] = max(
+                    per_model.get(r["model_name"], -np.inf),
+                    r["mean_test_score"],
+                )
+            print(per_model)
+
+        if self.refit:
+            bestm = models[self.best_model_index_]
+            est = _clone(bestm[1])
+            if hasattr(est, "sc"):
+                est.sc = None
+            est.set_params(**self.best_params_)
+            est.fit(X, y, **fit_params)
+            self.best_estimator_ = est
+
+        del self.sc
+        self.sc = None
+        return self
+
+    # delegation ------------------------------------------------------- #
+    def _fitted(self):
+        if not self.refit:
+            raise AttributeError(
+                "refit=False: only best_params_ is available"
+            )
+        return self.best_estimator_
+
+    @property
+    def classes_(self):
+        return self._fitted().classes_
+
+    def predict(self, X):
+        return self._fitted().predict(X)
+
+    def predict_proba(self, X):
+        return self._fitted().predict_proba(X)
+
+    def predict_log_proba(self, X):
+        return self._fitted().predict_log_proba(X)
+
+    def decision_function(self, X):
+        return self._fitted().decision_function(X)
+
+    def transform(self, X):
+        return self._fitted().transform(X)
+
+    def inverse_transform(self, Xt):
+        return self._fitted().inverse_transform(Xt)
+
+
 class DistRandomizedSearchCV(DistBaseSearchCV):
     """Distributed randomized search (reference search.py:648-714)."""
 

@@ -1,0 +1,194 @@
+"""
+Pipeline-safe preprocessing transformers (reference: skdist/preprocessing.py).
+
+Same transformer set and semantics as the reference, implemented directly
+as Transformer classes (the reference wrapped each in a
+``FunctionTransformer``, preprocessing.py:42-49).
+"""
+
+import warnings
+
+import numpy as np
+import pandas as pd
+import scipy.sparse as sparse
+from sklearn import feature_selection
+from sklearn.base import BaseEstimator, TransformerMixin
+from sklearn.feature_extraction.text import HashingVectorizer
+from sklearn.preprocessing import LabelEncoder, MultiLabelBinarizer, normalize
+
+__all__ = [
+    "SelectField", "DenseTransformer", "SparseTransformer", "FeatureCast",
+    "ImputeNull", "LabelEncoderPipe", "SelectorMem",
+    "HashingVectorizerChunked", "MultihotEncoder",
+]
+
+
+class _StatelessTransformer(TransformerMixin, BaseEstimator):
+    """Base for transformers whose fit is a no-op."""
+
+    def fit(self, X, y=None):
+        return self
+
+
+class SelectField(_StatelessTransformer):
+    """Select column(s) from a pandas DataFrame as a numpy array
+    (reference preprocessing.py:77-94).
+
+    Args:
+        cols: list of columns to select (None -> all).
+        single_dimension: with exactly one column, return it 1-D.
+    """
+
+    def __init__(self, cols=None, single_dimension=False):
+        self.cols = cols
+        self.single_dimension = single_dimension
+
+    def transform(self, X, y=None):
+        if self.cols is None:
+            return X.values
+        if len(self.cols) == 1 and self.single_dimension:
+            return X[self.cols[0]].values
+        return X[self.cols].values
+
+
+class DenseTransformer(_StatelessTransformer):
+    """Densify sparse input (reference preprocessing.py:105-109)."""
+
+    def transform(self, X, y=None):
+        return X.todense() if sparse.issparse(X) else X
+
+
+class SparseTransformer(_StatelessTransformer):
+    """CSR-sparsify dense input (reference preprocessing.py:120-124)."""
+
+    def transform(self, X, y=None):
+        return X if sparse.issparse(X) else sparse.csr_matrix(X)
+
+
+class FeatureCast(_StatelessTransformer):
+    """Cast feature dtype (reference preprocessing.py:143-154)."""
+
+    def __init__(self, cast_type=None):
+        self.cast_type = cast_type
+
+    def transform(self, X, y=None):
+        return X if self.cast_type is None else X.astype(self.cast_type)
+
+
+class ImputeNull(_StatelessTransformer):
+    """Fill nulls (pd.isnull semantics) with a constant
+    (reference preprocessing.py:175-186)."""
+
+    def __init__(self, impute_val=None):
+        self.impute_val = impute_val
+
+    def transform(self, X, y=None):
+        if self.impute_val is None:
+            return X
+        X = np.asarray(X, dtype=object) if not isinstance(
+            X, np.ndarray
+        ) else X
+        out = X.copy()
+        out[pd.isnull(out)] = self.impute_val
+        return out
+
+
+class LabelEncoderPipe(TransformerMixin, BaseEstimator):
+    """LabelEncoder usable inside a Pipeline/FeatureUnion
+    (reference preprocessing.py:189-203): output is a column vector."""
+
+    def fit(self, X, y=None):
+        self.le = LabelEncoder().fit(X)
+        return self
+
+    def transform(self, X, y=None):
+        return self.le.transform(X).reshape(-1, 1)
+
+
+_SELECTORS = {
+    "fpr": feature_selection.SelectFpr,
+    "fdr": feature_selection.SelectFdr,
+    "fwe": feature_selection.SelectFwe,
+    "kbest": feature_selection.SelectKBest,
+    "percentile": feature_selection.SelectPercentile,
+}
+
+
+class SelectorMem(TransformerMixin, BaseEstimator):
+    """Univariate feature selector that stores only the smaller of
+    {boolean mask, index list} (reference preprocessing.py:206-261)."""
+
+    def __init__(self, selector="fpr", score_func=feature_selection.f_classif,
+                 threshold=0.05):
+        self.selector = selector
+        self.score_func = score_func
+        self.threshold = threshold
+
+    def fit(self, X, y=None):
+        name = self.selector.lower()
+        cls = _SELECTORS[name]
+        if name == "kbest":
+            sel = cls(self.score_func, k=self.threshold)
+        elif name == "percentile":
+            sel = cls(self.score_func, percentile=self.threshold)
+        else:
+            sel = cls(self.score_func, alpha=self.threshold)
+        sel.fit(X, y)
+        idx = sel.get_support(indices=True)
+        boolean = sel.get_support(indices=False)
+        self.mask = idx if boolean.nbytes > np.asarray(idx).nbytes else boolean
+        return self
+
+    def transform(self, X, y=None):
+        return X[:, self.mask]
+
+
+class HashingVectorizerChunked(HashingVectorizer):
+    """HashingVectorizer with chunked transform to bound peak memory
+    (reference preprocessing.py:264-310)."""
+
+    def __init__(self, chunksize=100000, **kwargs):
+        self.chunksize = chunksize
+        HashingVectorizer.__init__(self, **kwargs)
+
+    def transform(self, X):
+        if isinstance(X, str):
+            raise ValueError(
+                "Iterable over raw text documents expected, string object "
+                "received."
+            )
+        if self.chunksize is None or len(X) < self.chunksize:
+            return self._transform_chunk(X)
+        return sparse.vstack(
+            [
+                self._transform_chunk(X[i : i + self.chunksize])
+                for i in range(0, len(X), self.chunksize)
+            ]
+        )
+
+    def _transform_chunk(self, docs):
+        analyzer = self.build_analyzer()
+        out = self._get_hasher().transform(analyzer(d) for d in docs)
+        if self.binary:
+            out.data.fill(1)
+        if self.norm is not None:
+            out = normalize(out, norm=self.norm, copy=False)
+        return out
+
+
+class MultihotEncoder(TransformerMixin, BaseEstimator):
+    """Pipeline-safe MultiLabelBinarizer
+    (reference preprocessing.py:313-339)."""
+
+    def __init__(self, sparse_output=False):
+        self.sparse_output = sparse_output
+
+    def fit(self, X, y=None):
+        self.transformer = MultiLabelBinarizer().fit(X)
+        return self
+
+    def transform(self, X, y=None):
+        with warnings.catch_warnings():
+            warnings.simplefilter("ignore")
+            out = self.transformer.transform(X)
+        return sparse.csr_matrix(out) if self.sparse_output else out

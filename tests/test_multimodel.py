@@ -1,0 +1,56 @@
+"""DistMultiModelSearch tests (reference search.py:717-908)."""
+
+import numpy as np
+from scipy.stats import uniform
+from sklearn.datasets import load_breast_cancer
+from sklearn.ensemble import RandomForestClassifier
+from sklearn.linear_model import LogisticRegression as SkLogReg
+
+from skdist_amd import Cluster
+from skdist_amd.distribute.search import DistMultiModelSearch
+
+
+def _models():
+    return [
+        ("logreg", SkLogReg(solver="liblinear"), {"C": [0.1, 1.0, 10.0]}),
+        ("rf", RandomForestClassifier(n_estimators=10, random_state=0),
+         {"max_depth": [2, 4, 8]}, 2),
+    ]
+
+
+def test_multimodel_local():
+    X, y = load_breast_cancer(return_X_y=True)
+    mm = DistMultiModelSearch(
+        _models(), n=3, cv=3, random_state=0, scoring="roc_auc"
+    )
+    mm.fit(X, y)
+    assert mm.best_score_ > 0.95
+    assert mm.best_model_name_ in ("logreg", "rf")
+    assert mm.worst_score_ <= mm.best_score_
+    r = mm.cv_results_
+    assert len(r["params"]) == len(r["mean_test_score"])
+    assert set(r["model_name"]) == {"logreg", "rf"}
+    # rf capped at n=2 samples, logreg n=3
+    assert r["model_name"].count("rf") == 2
+    preds = mm.predict(X[:5])
+    assert preds.shape == (5,)
+
+
+def test_multimodel_cluster():
+    X, y = load_breast_cancer(return_X_y=True)
+    mm = DistMultiModelSearch(
+        _models(), n=2, cv=3, random_state=0, sc=Cluster()
+    )
+    mm.fit(X, y)
+    assert mm.sc is None
+    assert mm.best_score_ > 0.9
+
+
+def test_multimodel_continuous_dists():
+    X, y = load_breast_cancer(return_X_y=True)
+    mm = DistMultiModelSearch(
+        [("lr", SkLogReg(solver="liblinear"), {"C": uniform(0.1, 5)})],
+        n=3, cv=3, random_state=7,
+    )
+    mm.fit(X, y)
+    assert len(mm.cv_results_["params"]) == 3
