@@ -1,0 +1,57 @@
+"""
+Batched inference tests (reference mirror:
+skdist/distribute/tests/test_predict.py + test_spark.py:136-162).
+"""
+
+import numpy as np
+import pandas as pd
+from sklearn.datasets import load_breast_cancer
+from sklearn.linear_model import LogisticRegression as SkLogReg
+
+from skdist_amd.distribute.predict import DistPredictor, get_prediction_fn
+
+
+def _fitted():
+    X, y = load_breast_cancer(return_X_y=True)
+    return SkLogReg(solver="liblinear").fit(X, y), X, y
+
+
+def test_prediction_fn_numpy():
+    model, X, y = _fitted()
+    fn = get_prediction_fn(model, method="predict", feature_type="numpy")
+    cols = [X[:, i] for i in range(X.shape[1])]
+    preds = fn(*cols)
+    assert (preds == model.predict(X)).all()
+
+
+def test_prediction_fn_proba():
+    model, X, y = _fitted()
+    fn = get_prediction_fn(model, method="predict_proba",
+                           feature_type="numpy")
+    cols = [X[:, i] for i in range(X.shape[1])]
+    p = fn(*cols)
+    assert p.shape == (len(y), 2)
+    assert np.allclose(p, model.predict_proba(X))
+
+
+def test_prediction_fn_pandas():
+    model, X, y = _fitted()
+    names = [f"f{i}" for i in range(X.shape[1])]
+
+    class PdModel:
+        def predict(self, df):
+            assert isinstance(df, pd.DataFrame)
+            return model.predict(df.values)
+
+    fn = get_prediction_fn(PdModel(), feature_type="pandas", names=names)
+    preds = fn(*[X[:, i] for i in range(X.shape[1])])
+    assert (preds == model.predict(X)).all()
+
+
+def test_dist_predictor_local_chunked():
+    model, X, y = _fitted()
+    pred = DistPredictor(model, chunk_rows=100)
+    out = pred.predict(X)
+    assert (out == model.predict(X)).all()
+    proba = DistPredictor(model, method="predict_proba", chunk_rows=64)
+    assert np.allclose(proba.predict(X), model.predict_proba(X))
