@@ -160,3 +160,27 @@ def test_gpu_cluster_requires_gpu_flag():
 
     c = Cluster(require_gpu=True)
     assert c.device.type == "cuda"
+
+
+def test_ovr_ovo_batched_gpu():
+    from sklearn.datasets import load_digits
+    from sklearn.metrics import accuracy_score
+
+    from skdist_amd import Cluster
+    from skdist_amd.distribute.multiclass import (
+        DistOneVsOneClassifier,
+        DistOneVsRestClassifier,
+    )
+    from skdist_amd.models import LogisticRegression
+
+    X, y = load_digits(return_X_y=True)
+    est = LogisticRegression(epochs=20, random_state=0)
+    ovr = DistOneVsRestClassifier(est, sc=Cluster(require_gpu=True))
+    ovr.fit(X, y)
+    assert accuracy_score(y, ovr.predict(X)) > 0.93
+    assert len(ovr.estimators_) == 10
+
+    ovo = DistOneVsOneClassifier(est, sc=Cluster(require_gpu=True))
+    ovo.fit(X, y)
+    assert accuracy_score(y, ovo.predict(X)) > 0.93
+    assert len(ovo.estimators_) == 45
