@@ -86,6 +86,19 @@ def _stale(obj, deps):
     return any(os.path.getmtime(d) > omt for d in deps)
 
 
+def extension_is_stale():
+    """True when any csrc file is newer than the built .so (a stale .so
+    on the GPU box runs OLD kernels — fail loudly instead)."""
+    if not os.path.exists(OUT):
+        return True
+    omt = os.path.getmtime(OUT)
+    for fn in os.listdir(CSRC):
+        if fn.endswith((".hip", ".cpp", ".h")):
+            if os.path.getmtime(os.path.join(CSRC, fn)) > omt:
+                return True
+    return False
+
+
 if __name__ == "__main__":
     build()
     sys.exit(0)

@@ -21,3 +21,15 @@ def pytest_collection_modifyitems(config, items):
     for item in items:
         if "gpu" in item.keywords:
             item.add_marker(skip)
+
+
+def pytest_sessionstart(session):
+    # a stale in-tree extension would ship OLD kernels to the GPU box;
+    # fail the CPU suite early instead
+    from skdist_amd.ops.build import extension_is_stale
+
+    if extension_is_stale():
+        raise RuntimeError(
+            "skdist_amd/ops/_skdist_hip.so is stale or missing — run "
+            "`python -m skdist_amd.ops.build`"
+        )
