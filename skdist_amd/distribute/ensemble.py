@@ -21,6 +21,7 @@ into the same classes as a device fast path.
 
 import numpy as np
 import scipy.sparse as sp
+import torch
 from sklearn.base import BaseEstimator, ClassifierMixin, RegressorMixin, TransformerMixin
 from sklearn.preprocessing import OneHotEncoder
 from sklearn.tree import (
@@ -48,6 +49,10 @@ def _bootstrap_weights(seed, n, base_weight=None):
     return counts
 
 
+def _oob_indices_from_seed(seed, n):
+    return np.flatnonzero(_bootstrap_weights(seed, n) == 0)
+
+
 def _resolve_max_features(max_features, is_classifier):
     # the reference era's 'auto' (sqrt for clf, n for reg); modern sklearn
     # rejects 'auto'
@@ -72,12 +77,27 @@ def _fit_one_tree(tree_proto, X, y, seed, bootstrap, sample_weight,
 
 class DistBaseForest(BaseEstimator):
     """Shared fan-out machinery for all forest classes
-    (reference ensemble.py:154-340)."""
+    (reference ensemble.py:154-340).
+
+    Two per-tree engines behind the same fan-out:
+      * CPU path (``sc=None`` or unsupported params): one sklearn tree per
+        task, like the reference worker (ensemble.py:68-109);
+      * device path (``sc`` on a GPU): the batched HIP histogram builder
+        (skdist_amd.models.forest) grows each rank's shard of trees
+        level-synchronously against the HBM-resident binned data — the
+        trees gathered back are host-side ``HistTree`` arrays that pickle
+        and predict exactly like the CPU ones.
+    """
 
     _is_classifier = False
 
     def _tree_proto(self):
         raise NotImplementedError
+
+    def _device_spec(self):
+        """dict(criterion=..., extra=..., max_features=...) when this
+        class supports the batched HIP builder, else None."""
+        return None
 
     def fit(self, X, y, sample_weight=None):
         _check_estimator(self, verbose=bool(self.verbose))
@@ -93,10 +113,10 @@ class DistBaseForest(BaseEstimator):
 
         rnd = check_random_state(self.random_state)
         seeds = rnd.randint(MAX_SEED, size=self.n_estimators)
-        proto = self._tree_proto()
 
         if not self.warm_start or not hasattr(self, "estimators_"):
             self.estimators_ = []
+            self._oob_idx = []
         n_more = self.n_estimators - len(self.estimators_)
         if n_more < 0:
             raise ValueError(
@@ -105,19 +125,34 @@ class DistBaseForest(BaseEstimator):
             )
         seeds = seeds[:n_more]
 
-        def task_fn(task):
-            i, seed = task
-            return i, _fit_one_tree(
-                proto, X, y, int(seed), self.bootstrap, sample_weight
-            )
-
-        tasks = list(enumerate(seeds))
-        if sc is None:
-            results = run_local_tasks(task_fn, tasks, n_jobs=self.n_jobs)
+        if self._device_fit_ok(sc, X, sample_weight):
+            results = self._fit_trees_device(sc, X, y, seeds)
+            self.estimators_.extend(r[0] for r in results)
+            if self.oob_score:
+                self._oob_idx.extend(r[1] for r in results)
         else:
-            results = sc.run_tasks(task_fn, tasks)
-        results.sort(key=lambda t: t[0])
-        self.estimators_.extend(r[1] for r in results)
+            proto = self._tree_proto()
+
+            def task_fn(task):
+                i, seed = task
+                return i, _fit_one_tree(
+                    proto, X, y, int(seed), self.bootstrap, sample_weight
+                )
+
+            tasks = list(enumerate(seeds))
+            if sc is None:
+                results = run_local_tasks(
+                    task_fn, tasks, n_jobs=self.n_jobs)
+            else:
+                results = sc.run_tasks(task_fn, tasks)
+            results.sort(key=lambda t: t[0])
+            self.estimators_.extend(r[1] for r in results)
+            if self.oob_score:
+                self._oob_idx.extend(
+                    _oob_indices_from_seed(int(s), X.shape[0])
+                    if self.bootstrap else np.arange(X.shape[0])
+                    for s in seeds
+                )
         self._seeds = [int(s) for s in seeds]
 
         if self.oob_score:
@@ -126,21 +161,77 @@ class DistBaseForest(BaseEstimator):
         return self
 
     # ------------------------------------------------------------------ #
+    def _device_fit_ok(self, sc, X, sample_weight):
+        spec = self._device_spec()
+        if spec is None or sc is None:
+            return False
+        dev = getattr(sc, "device", None)
+        if dev is None or dev.type != "cuda":
+            return False
+        if sp.issparse(X) or sample_weight is not None:
+            return False
+        if getattr(self, "max_leaf_nodes", None) is not None:
+            return False
+        if getattr(self, "min_weight_fraction_leaf", 0.0):
+            return False
+        if getattr(self, "class_weight", None) is not None:
+            return False
+        if self._is_classifier:
+            from ..models.forest import MAX_DEVICE_CLASSES
+
+            if self.n_classes_ > MAX_DEVICE_CLASSES:
+                return False
+        return True
+
+    def _fit_trees_device(self, sc, X, y, seeds):
+        """Build this rank's shard of trees with the HIP histogram
+        builder; all-gather → ordered [(tree, oob_idx)] list."""
+        from ..models.forest import BinnedDataset, ForestBuilder
+
+        spec = self._device_spec()
+        n = X.shape[0]
+        mss = self.min_samples_split
+        mss = mss if mss >= 2 else int(np.ceil(mss * n))
+        msl = self.min_samples_leaf
+        msl = msl if msl >= 1 else int(np.ceil(msl * n))
+        ds = BinnedDataset(
+            X, y, sc.device, is_cls=self._is_classifier,
+            classes=self.classes_ if self._is_classifier else None,
+        )
+        builder = ForestBuilder(
+            ds, spec["criterion"], max_depth=self.max_depth,
+            min_samples_split=mss, min_samples_leaf=msl,
+            min_impurity_decrease=self.min_impurity_decrease,
+            max_features=spec["max_features"], extra_mode=spec["extra"],
+            bootstrap=self.bootstrap,
+        )
+        mine = sc.shard_indices(len(seeds))
+        my_seeds = [int(seeds[i]) for i in mine]
+        trees = builder.build(my_seeds)
+        if self.oob_score and self.bootstrap:
+            w = builder.make_weights(my_seeds)
+            oob = [
+                torch.nonzero(w[k] == 0).flatten().cpu().numpy()
+                for k in range(len(my_seeds))
+            ]
+        else:
+            oob = [np.arange(n)] * len(my_seeds)
+        local = {i: (trees[k], oob[k]) for k, i in enumerate(mine)}
+        return sc.gather_task_results(local, len(seeds))
+
+    # ------------------------------------------------------------------ #
     def _compute_oob(self, X, y):
         """Real OOB scoring (the reference stubs this out,
-        ensemble.py:338-340)."""
+        ensemble.py:338-340).  OOB rows per tree were recorded at fit
+        time (``_oob_idx``) so the CPU numpy bootstrap and the device
+        torch-RNG bootstrap both score correctly."""
         n = X.shape[0]
         if self._is_classifier:
             agg = np.zeros((n, self.n_classes_))
         else:
             agg = np.zeros(n)
         cnt = np.zeros(n)
-        for seed, tree in zip(self._seeds, self.estimators_):
-            counts = _bootstrap_weights(seed, n) if self.bootstrap else None
-            oob = (
-                np.flatnonzero(counts == 0) if counts is not None
-                else np.arange(n)
-            )
+        for oob, tree in zip(self._oob_idx, self.estimators_):
             if len(oob) == 0:
                 continue
             if self._is_classifier:
@@ -259,6 +350,12 @@ class DistRandomForestClassifier(_ForestClassifierMixin, DistBaseForest):
             max_features=_resolve_max_features(self.max_features, True),
         )
 
+    def _device_spec(self):
+        if self.criterion not in ("gini", "entropy", "log_loss"):
+            return None
+        return {"criterion": self.criterion, "extra": False,
+                "max_features": _resolve_max_features(self.max_features, True)}
+
 
 class DistExtraTreesClassifier(_ForestClassifierMixin, DistBaseForest):
     """Distributed extra-trees classifier (reference ensemble.py:424-480)."""
@@ -288,6 +385,12 @@ class DistExtraTreesClassifier(_ForestClassifierMixin, DistBaseForest):
             max_features=_resolve_max_features(self.max_features, True),
         )
 
+    def _device_spec(self):
+        if self.criterion not in ("gini", "entropy", "log_loss"):
+            return None
+        return {"criterion": self.criterion, "extra": True,
+                "max_features": _resolve_max_features(self.max_features, True)}
+
 
 class DistRandomForestRegressor(_ForestRegressorMixin, DistBaseForest):
     """Distributed random forest regressor (reference ensemble.py:505-559)."""
@@ -316,6 +419,12 @@ class DistRandomForestRegressor(_ForestRegressorMixin, DistBaseForest):
             max_features=_resolve_max_features(self.max_features, False),
         )
 
+    def _device_spec(self):
+        if self.criterion not in ("squared_error", "mse"):
+            return None
+        return {"criterion": "squared_error", "extra": False,
+                "max_features": _resolve_max_features(self.max_features, False)}
+
 
 class DistExtraTreesRegressor(_ForestRegressorMixin, DistBaseForest):
     """Distributed extra-trees regressor (reference ensemble.py:562-616)."""
@@ -343,6 +452,12 @@ class DistExtraTreesRegressor(_ForestRegressorMixin, DistBaseForest):
             **{k: getattr(self, k) for k in _COMMON_TREE_PARAMS},
             max_features=_resolve_max_features(self.max_features, False),
         )
+
+    def _device_spec(self):
+        if self.criterion not in ("squared_error", "mse"):
+            return None
+        return {"criterion": "squared_error", "extra": True,
+                "max_features": _resolve_max_features(self.max_features, False)}
 
 
 class DistRandomTreesEmbedding(TransformerMixin, DistBaseForest):
@@ -375,6 +490,10 @@ class DistRandomTreesEmbedding(TransformerMixin, DistBaseForest):
             max_features=1, max_leaf_nodes=self.max_leaf_nodes,
             min_impurity_decrease=self.min_impurity_decrease,
         )
+
+    def _device_spec(self):
+        return {"criterion": "squared_error", "extra": True,
+                "max_features": 1}
 
     def fit(self, X, y=None, sample_weight=None):
         self.fit_transform(X, y, sample_weight=sample_weight)

@@ -93,7 +93,7 @@ class DistPredictor:
         return np.concatenate(boxes, axis=0)
 
     def _predict_local(self, X):
-        fn = getattr(self.model, self.method)
+        fn = self._device_fn() or getattr(self.model, self.method)
         n = X.shape[0]
         if n <= self.chunk_rows:
             return np.asarray(fn(X))
@@ -101,3 +101,26 @@ class DistPredictor:
         for lo in range(0, n, self.chunk_rows):
             outs.append(np.asarray(fn(X[lo : lo + self.chunk_rows])))
         return np.concatenate(outs, axis=0)
+
+    def _device_fn(self):
+        """GPU fast path: HIP-fitted forests score through the batched
+        traversal kernel instead of the host loop (the device analog of
+        the reference's executor-side predict, predict.py:160-178)."""
+        if self.method not in ("predict", "predict_proba"):
+            return None
+        import torch
+
+        if not torch.cuda.is_available():
+            return None
+        if getattr(self, "_flat", None) is None:
+            from ..models.forest import flat_forest_for
+
+            device = (
+                self.sc.device
+                if getattr(self.sc, "device", None) is not None
+                else "cuda"
+            )
+            self._flat = flat_forest_for(self.model, device) or False
+        if self._flat is False:
+            return None
+        return getattr(self._flat, self.method)

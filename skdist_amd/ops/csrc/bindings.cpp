@@ -108,7 +108,149 @@ void sgd_epoch(torch::Tensor Xs, torch::Tensor XsT, torch::Tensor GT,
 
 }  // namespace
 
+// ------------------------------------------------------------------ //
+// histogram tree builder (tree_kernels.hip)
+// ------------------------------------------------------------------ //
+
+extern "C" hipError_t skdist_tree_hist(
+    const void* codes, const void* y_int, const void* y_f,
+    const void* weights, const void* sample_idx, const void* chunks,
+    void* hist, long long n, int f, int nbins, int S, int is_cls, int fg,
+    int n_chunks, hipStream_t stream);
+extern "C" hipError_t skdist_tree_split(
+    const void* hist, const void* node_seed, int n_frontier, int f,
+    int nbins, int S, int is_cls, int crit, int m_features, int extra_mode,
+    float min_leaf_w, void* out_feat, void* out_bin, void* out_wl,
+    void* out_gain, void* out_imp, void* out_stats, void* out_lstats,
+    hipStream_t stream);
+extern "C" hipError_t skdist_part_count(
+    const void* codes, const void* sample_idx, const void* chunks,
+    const void* split_feat, const void* split_bin, long long n,
+    int n_chunks, void* out_counts, hipStream_t stream);
+extern "C" hipError_t skdist_part_scatter(
+    const void* codes, const void* sample_idx_in, const void* chunks,
+    const void* split_feat, const void* split_bin, const void* left_base,
+    const void* right_base, long long n, int n_chunks,
+    void* sample_idx_out, hipStream_t stream);
+extern "C" hipError_t skdist_forest_predict(
+    const void* X, const void* feat, const void* thr, const void* left,
+    const void* right, const void* roots, const void* values, void* out,
+    long long rows, int f, int n_trees, int vs, hipStream_t stream);
+extern "C" hipError_t skdist_forest_apply(
+    const void* X, const void* feat, const void* thr, const void* left,
+    const void* right, const void* roots, void* out_leaf, long long rows,
+    int f, int n_trees, hipStream_t stream);
+
+namespace {
+
+void tree_hist(torch::Tensor codes, torch::Tensor y_int, torch::Tensor y_f,
+               torch::Tensor weights, torch::Tensor sample_idx,
+               torch::Tensor chunks, torch::Tensor hist, int64_t n,
+               int64_t f, int64_t nbins, int64_t S, int64_t is_cls,
+               int64_t fg) {
+    for (auto* t : {&codes, &weights, &sample_idx, &chunks, &hist}) {
+        CHECK_DEV(*t);
+        CHECK_CONT(*t);
+    }
+    TORCH_CHECK(codes.scalar_type() == torch::kUInt8, "codes must be u8");
+    TORCH_CHECK(chunks.size(1) == 4, "chunks must be [n_chunks, 4]");
+    auto stream = c10::hip::getCurrentHIPStream().stream();
+    hipError_t err = skdist_tree_hist(
+        codes.data_ptr(), is_cls ? y_int.data_ptr() : nullptr,
+        is_cls ? nullptr : y_f.data_ptr(), weights.data_ptr(),
+        sample_idx.data_ptr(), chunks.data_ptr(), hist.data_ptr(), n,
+        (int)f, (int)nbins, (int)S, (int)is_cls, (int)fg,
+        (int)chunks.size(0), stream);
+    TORCH_CHECK(err == hipSuccess, "tree_hist: ", hipGetErrorString(err));
+}
+
+void tree_split(torch::Tensor hist, torch::Tensor node_seed, int64_t f,
+                int64_t nbins, int64_t S, int64_t is_cls, int64_t crit,
+                int64_t m_features, int64_t extra_mode, double min_leaf_w,
+                torch::Tensor out_feat, torch::Tensor out_bin,
+                torch::Tensor out_wl, torch::Tensor out_gain,
+                torch::Tensor out_imp, torch::Tensor out_stats,
+                torch::Tensor out_lstats) {
+    CHECK_DEV(hist);
+    CHECK_CONT(hist);
+    TORCH_CHECK(S <= 32, "tree_split supports <= 32 stats");
+    auto stream = c10::hip::getCurrentHIPStream().stream();
+    hipError_t err = skdist_tree_split(
+        hist.data_ptr(), node_seed.data_ptr(), (int)node_seed.size(0),
+        (int)f, (int)nbins, (int)S, (int)is_cls, (int)crit,
+        (int)m_features, (int)extra_mode, (float)min_leaf_w,
+        out_feat.data_ptr(), out_bin.data_ptr(), out_wl.data_ptr(),
+        out_gain.data_ptr(), out_imp.data_ptr(), out_stats.data_ptr(),
+        out_lstats.data_ptr(), stream);
+    TORCH_CHECK(err == hipSuccess, "tree_split: ", hipGetErrorString(err));
+}
+
+void part_count(torch::Tensor codes, torch::Tensor sample_idx,
+                torch::Tensor chunks, torch::Tensor split_feat,
+                torch::Tensor split_bin, int64_t n,
+                torch::Tensor out_counts) {
+    auto stream = c10::hip::getCurrentHIPStream().stream();
+    hipError_t err = skdist_part_count(
+        codes.data_ptr(), sample_idx.data_ptr(), chunks.data_ptr(),
+        split_feat.data_ptr(), split_bin.data_ptr(), n,
+        (int)chunks.size(0), out_counts.data_ptr(), stream);
+    TORCH_CHECK(err == hipSuccess, "part_count: ", hipGetErrorString(err));
+}
+
+void part_scatter(torch::Tensor codes, torch::Tensor sample_idx_in,
+                  torch::Tensor chunks, torch::Tensor split_feat,
+                  torch::Tensor split_bin, torch::Tensor left_base,
+                  torch::Tensor right_base, int64_t n,
+                  torch::Tensor sample_idx_out) {
+    auto stream = c10::hip::getCurrentHIPStream().stream();
+    hipError_t err = skdist_part_scatter(
+        codes.data_ptr(), sample_idx_in.data_ptr(), chunks.data_ptr(),
+        split_feat.data_ptr(), split_bin.data_ptr(), left_base.data_ptr(),
+        right_base.data_ptr(), n, (int)chunks.size(0),
+        sample_idx_out.data_ptr(), stream);
+    TORCH_CHECK(err == hipSuccess, "part_scatter: ",
+                hipGetErrorString(err));
+}
+
+void forest_predict(torch::Tensor X, torch::Tensor feat, torch::Tensor thr,
+                    torch::Tensor left, torch::Tensor right,
+                    torch::Tensor roots, torch::Tensor values,
+                    torch::Tensor out) {
+    for (auto* t : {&X, &thr, &values, &out}) {
+        CHECK_DEV(*t);
+        CHECK_CONT(*t);
+    }
+    auto stream = c10::hip::getCurrentHIPStream().stream();
+    hipError_t err = skdist_forest_predict(
+        X.data_ptr(), feat.data_ptr(), thr.data_ptr(), left.data_ptr(),
+        right.data_ptr(), roots.data_ptr(), values.data_ptr(),
+        out.data_ptr(), X.size(0), (int)X.size(1), (int)roots.size(0),
+        (int)values.size(1), stream);
+    TORCH_CHECK(err == hipSuccess, "forest_predict: ",
+                hipGetErrorString(err));
+}
+
+void forest_apply(torch::Tensor X, torch::Tensor feat, torch::Tensor thr,
+                  torch::Tensor left, torch::Tensor right,
+                  torch::Tensor roots, torch::Tensor out_leaf) {
+    auto stream = c10::hip::getCurrentHIPStream().stream();
+    hipError_t err = skdist_forest_apply(
+        X.data_ptr(), feat.data_ptr(), thr.data_ptr(), left.data_ptr(),
+        right.data_ptr(), roots.data_ptr(), out_leaf.data_ptr(), X.size(0),
+        (int)X.size(1), (int)roots.size(0), stream);
+    TORCH_CHECK(err == hipSuccess, "forest_apply: ",
+                hipGetErrorString(err));
+}
+
+}  // namespace
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("sgd_step", &sgd_step, "fused batched SGD step (K1+K2+K3)");
     m.def("sgd_epoch", &sgd_epoch, "one epoch of fused SGD steps");
+    m.def("tree_hist", &tree_hist, "per-(node,feature,bin) stats");
+    m.def("tree_split", &tree_split, "best split per frontier node");
+    m.def("part_count", &part_count, "partition phase A: left counts");
+    m.def("part_scatter", &part_scatter, "partition phase B: scatter");
+    m.def("forest_predict", &forest_predict, "batched forest inference");
+    m.def("forest_apply", &forest_apply, "leaf ids per (row, tree)");
 }
