@@ -42,6 +42,7 @@ def main():
     ap.add_argument("--rows", type=int, default=1_000_000)
     ap.add_argument("--features", type=int, default=256)
     ap.add_argument("--epochs", type=int, default=20)
+    ap.add_argument("--batch-size", type=int, default=8192)
     args = ap.parse_args()
 
     import torch
@@ -64,7 +65,7 @@ def main():
 
     def one_step():
         est = LogisticRegression(
-            epochs=args.epochs, batch_size=8192, random_state=0
+            epochs=args.epochs, batch_size=args.batch_size, random_state=0
         )
         gs = DistGridSearchCV(
             est, grid, cv=args.folds, scoring="accuracy", sc=cluster

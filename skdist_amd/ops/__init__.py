@@ -46,6 +46,82 @@ def require_hip():
     return _ext
 
 
+def hash_vectorize(docs, n_features=2 ** 20, analyzer="word",
+                   ngram_range=(1, 1), alternate_sign=True, binary=False,
+                   norm="l2", lowercase=True, dtype="float64",
+                   device="cuda"):
+    """Device hashing vectorizer: tokenize + murmur3 + CSR, sklearn-exact
+    for ASCII input (kernel: csrc/hash_kernels.hip; reference analog:
+    sklearn HashingVectorizer inside skdist/preprocessing.py:264-310).
+
+    Documents are packed into ONE byte buffer uploaded to HBM; the kernel
+    emits (doc*F + feature, ±1) COO pairs; torch sorts + segment-sums
+    them and the result returns as a scipy CSR (same dtype/norm contract
+    as sklearn's transform).
+    """
+    import numpy as np
+    import scipy.sparse as sp
+
+    ext = require_hip()
+    mode = {"word": 0, "char_wb": 1}[analyzer]
+    min_n, max_n = ngram_range
+    enc = [
+        (d.lower() if lowercase else d).encode("utf-8") for d in docs
+    ]
+    lens = np.fromiter((len(b) for b in enc), dtype=np.int64,
+                       count=len(enc))
+    off = np.zeros(len(enc) + 1, dtype=np.int64)
+    np.cumsum(lens, out=off[1:])
+    blob = np.frombuffer(b"".join(enc), dtype=np.uint8)
+    n_docs = len(enc)
+    dev = torch.device(device)
+    bytes_t = torch.as_tensor(blob, device=dev)
+    off_t = torch.as_tensor(off, device=dev)
+    orders = max_n - min_n + 1
+    if mode == 0:
+        cap = int(len(blob) // 3 + n_docs + 64) * orders
+    else:
+        cap = int(2 * len(blob) + 2 * n_docs + 64) * orders
+    while True:
+        keys = torch.empty(max(cap, 64), dtype=torch.int64, device=dev)
+        vals = torch.empty(max(cap, 64), dtype=torch.float32, device=dev)
+        ctr = torch.zeros(1, dtype=torch.int64, device=dev)
+        ext.hash_vectorize(bytes_t, off_t, mode, min_n, max_n,
+                           n_features, int(alternate_sign), keys, vals,
+                           ctr)
+        total = int(ctr.item())
+        if total <= cap:
+            break
+        cap = total  # exact size known now; one retry
+    keys = keys[:total]
+    vals = vals[:total]
+    if total:
+        keys, order = torch.sort(keys)
+        vals = vals[order]
+        uniq, inverse = torch.unique_consecutive(keys,
+                                                 return_inverse=True)
+        sums = torch.zeros(len(uniq), dtype=torch.float32, device=dev)
+        sums.scatter_add_(0, inverse, vals)
+        rows = (uniq // n_features).cpu().numpy()
+        cols = (uniq % n_features).cpu().numpy().astype(np.int64)
+        data = sums.cpu().numpy().astype(dtype)
+    else:
+        rows = np.empty(0, dtype=np.int64)
+        cols = np.empty(0, dtype=np.int64)
+        data = np.empty(0, dtype=dtype)
+    if binary:
+        data = np.ones_like(data)
+    indptr = np.zeros(n_docs + 1, dtype=np.int64)
+    np.cumsum(np.bincount(rows, minlength=n_docs), out=indptr[1:])
+    out = sp.csr_matrix((data, cols, indptr),
+                        shape=(n_docs, n_features))
+    if norm is not None:
+        from sklearn.preprocessing import normalize
+
+        out = normalize(out, norm=norm, copy=False)
+    return out
+
+
 def _pad_cols(t, ncols_pad, fill):
     import torch as _t
 

@@ -140,6 +140,10 @@ extern "C" hipError_t skdist_forest_apply(
     const void* X, const void* feat, const void* thr, const void* left,
     const void* right, const void* roots, void* out_leaf, long long rows,
     int f, int n_trees, hipStream_t stream);
+extern "C" hipError_t skdist_hash_vectorize(
+    const void* bytes, const void* doc_off, long long n_docs, int mode,
+    int min_n, int max_n, int n_features, int alt_sign, void* out_keys,
+    void* out_vals, void* ctr, long long cap, hipStream_t stream);
 
 namespace {
 
@@ -242,11 +246,33 @@ void forest_apply(torch::Tensor X, torch::Tensor feat, torch::Tensor thr,
                 hipGetErrorString(err));
 }
 
+void hash_vectorize(torch::Tensor bytes, torch::Tensor doc_off,
+                    int64_t mode, int64_t min_n, int64_t max_n,
+                    int64_t n_features, int64_t alt_sign,
+                    torch::Tensor out_keys, torch::Tensor out_vals,
+                    torch::Tensor ctr) {
+    for (auto* t : {&bytes, &doc_off, &out_keys, &out_vals, &ctr}) {
+        CHECK_DEV(*t);
+        CHECK_CONT(*t);
+    }
+    TORCH_CHECK(bytes.scalar_type() == torch::kUInt8, "bytes must be u8");
+    auto stream = c10::hip::getCurrentHIPStream().stream();
+    hipError_t err = skdist_hash_vectorize(
+        bytes.data_ptr(), doc_off.data_ptr(), doc_off.size(0) - 1,
+        (int)mode, (int)min_n, (int)max_n, (int)n_features, (int)alt_sign,
+        out_keys.data_ptr(), out_vals.data_ptr(), ctr.data_ptr(),
+        out_keys.size(0), stream);
+    TORCH_CHECK(err == hipSuccess, "hash_vectorize: ",
+                hipGetErrorString(err));
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("sgd_step", &sgd_step, "fused batched SGD step (K1+K2+K3)");
     m.def("sgd_epoch", &sgd_epoch, "one epoch of fused SGD steps");
+    m.def("hash_vectorize", &hash_vectorize,
+          "tokenize + murmur3 feature hashing -> COO pairs");
     m.def("tree_hist", &tree_hist, "per-(node,feature,bin) stats");
     m.def("tree_split", &tree_split, "best split per frontier node");
     m.def("part_count", &part_count, "partition phase A: left counts");

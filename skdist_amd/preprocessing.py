@@ -167,6 +167,10 @@ class HashingVectorizerChunked(HashingVectorizer):
         )
 
     def _transform_chunk(self, docs):
+        docs = list(docs)
+        out = self._try_device_transform(docs)
+        if out is not None:
+            return out
         analyzer = self.build_analyzer()
         out = self._get_hasher().transform(analyzer(d) for d in docs)
         if self.binary:
@@ -174,6 +178,39 @@ class HashingVectorizerChunked(HashingVectorizer):
         if self.norm is not None:
             out = normalize(out, norm=self.norm, copy=False)
         return out
+
+    def _try_device_transform(self, docs):
+        """HIP fast path (ops/csrc/hash_kernels.hip): tokenize + murmur3 +
+        CSR on the GPU, bit-exact with sklearn for ASCII docs under the
+        default word / char_wb analyzers; anything else (custom
+        tokenizers, stop words, non-ASCII) keeps the sklearn path."""
+        from .ops import hash_vectorize, hip_available
+
+        if not hip_available():
+            return None
+        if self.analyzer == "word":
+            if self.token_pattern != r"(?u)\b\w\w+\b":
+                return None
+            if self.ngram_range[1] > 8:
+                return None
+        elif self.analyzer == "char_wb":
+            if self.ngram_range[1] > 32:
+                return None
+        else:
+            return None
+        if (self.stop_words is not None or self.preprocessor is not None
+                or self.tokenizer is not None
+                or self.strip_accents is not None
+                or self.input != "content"):
+            return None
+        if not all(isinstance(d, str) and d.isascii() for d in docs):
+            return None
+        return hash_vectorize(
+            docs, n_features=self.n_features, analyzer=self.analyzer,
+            ngram_range=self.ngram_range,
+            alternate_sign=self.alternate_sign, binary=self.binary,
+            norm=self.norm, lowercase=self.lowercase, dtype=self.dtype,
+        )
 
 
 class MultihotEncoder(TransformerMixin, BaseEstimator):
