@@ -119,7 +119,7 @@ def main():
                 "n_samples": args.rows,
                 "n_features": args.features,
                 "epochs": args.epochs,
-                "global_batch": 8192,
+                "global_batch": args.batch_size,
                 "seq_len": None,
                 "parallelism": f"task-fanout-dp{world}",
                 "best_score": None if gs is None else gs.best_score_,
