@@ -4,8 +4,12 @@ Monkeypatches sync-bracketed timers around the flagship path's phases to
 find where non-kernel time goes; informs bench optimization only.
 """
 
+import os
+import sys
 import time
 from collections import defaultdict
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 import numpy as np
 import torch
