@@ -163,7 +163,7 @@ class DistBaseSearchCV(BaseEstimator, MetaEstimatorMixin, metaclass=ABCMeta):
             and not fit_params
         )
         out = None
-        refit_models = None
+        refit_fn = None
         if batched:
             from ..models.linear import FallbackToGeneric
 
@@ -180,7 +180,7 @@ class DistBaseSearchCV(BaseEstimator, MetaEstimatorMixin, metaclass=ABCMeta):
             except FallbackToGeneric:
                 out = None
         if isinstance(out, dict):
-            refit_models = out.get("refit_estimators")
+            refit_fn = out.get("refit_fn")
             out = out["tasks"]
         if out is None:
             out = self._run_task_grid(
@@ -204,13 +204,16 @@ class DistBaseSearchCV(BaseEstimator, MetaEstimatorMixin, metaclass=ABCMeta):
                 results[f"mean_test_{refit_metric}"][self.best_index_]
             )
         if self.refit:
-            if refit_models is not None and self.best_index_ in refit_models:
-                # batched path: the full-data model trained alongside the
-                # CV columns — refit is a column extraction, not a 2nd solve
-                self.best_estimator_ = refit_models[self.best_index_]
-                self.refit_time_ = getattr(
-                    self.best_estimator_, "fit_time_", 0.0
-                )
+            # batched path: the full-data model trained alongside the CV
+            # columns — refit is a lazy column extraction (only the
+            # winning candidate is materialized/gathered), not a 2nd solve
+            best = (
+                refit_fn(self.best_index_) if refit_fn is not None
+                else None
+            )
+            if best is not None:
+                self.best_estimator_ = best
+                self.refit_time_ = getattr(best, "fit_time_", 0.0)
             else:
                 best = _clone(base_estimator)
                 best.set_params(**self.best_params_)

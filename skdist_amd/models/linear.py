@@ -25,6 +25,7 @@ from ._sgd import (
     ColumnSpec,
     DeviceDataset,
     batched_scores,
+    batched_scores_by_fold,
     batched_sgd_fit,
 )
 
@@ -101,7 +102,8 @@ class _BatchedLinearBase(BaseEstimator):
         return 0 if rs is None else int(rs)
 
     def _store_fitted(self, ds, W, is_clf):
-        Wh = W.cpu().numpy()       # [fa(+pad), ncols]
+        Wh = W if isinstance(W, np.ndarray) else W.cpu().numpy()
+        # [fa(+pad), ncols]
         w_std = Wh[: ds.f].T       # [ncols, f] standardized space
         b_std = Wh[ds.intercept_row]
         w_raw = np.empty_like(w_std)
@@ -181,6 +183,7 @@ class _BatchedLinearBase(BaseEstimator):
             else list(range(n_cand))
         )
         col_fold, col_class, col_lr, col_l2, col_model = [], [], [], [], []
+        model_folds = []   # per model: its test fold, -2 for full-data
         local_tasks = []   # (task_id, model_idx, fold)
         full_models = {}   # cand_id -> model_idx of the full-data model
         model_idx = 0
@@ -197,6 +200,7 @@ class _BatchedLinearBase(BaseEstimator):
                     col_lr.append(lr)
                     col_l2.append(lam)
                     col_model.append(model_idx)
+                model_folds.append(fold)
                 local_tasks.append((ci * n_folds + fold, model_idx, fold))
                 model_idx += 1
             # one extra full-data model per candidate: trains alongside the
@@ -210,6 +214,7 @@ class _BatchedLinearBase(BaseEstimator):
                 col_lr.append(lr)
                 col_l2.append(lam)
                 col_model.append(model_idx)
+            model_folds.append(-2)
             full_models[ci] = model_idx
             model_idx += 1
 
@@ -229,9 +234,10 @@ class _BatchedLinearBase(BaseEstimator):
             )
             fit_time = time.perf_counter() - t0
             t1 = time.perf_counter()
-            scores = batched_scores(
-                ds, spec, W, self._loss, np.asarray(col_model),
-                n_models=model_idx, n_classes=n_classes, metric=metric,
+            scores = batched_scores_by_fold(
+                ds, W, np.asarray(model_folds),
+                np.asarray(col_class, dtype=np.int32),
+                n_classes=n_classes, metric=metric,
             )
             score_time = time.perf_counter() - t1
             per = fit_time / max(model_idx, 1)
@@ -244,34 +250,51 @@ class _BatchedLinearBase(BaseEstimator):
                     "fit_time": per,
                     "score_time": per_s,
                 }
-            # materialize a fitted estimator per candidate from the
-            # full-data columns (host numpy weights, sc-free, picklable)
+            # full-data columns stay as ONE host weight matrix; an
+            # estimator is materialized only for the candidate the search
+            # asks for (refit_fn below) — the reference refits the winner
+            # on the driver instead (search.py:543-550)
             ncols_per = len(cls)
-            Wh = W
-            for ci, mi in full_models.items():
-                cols = slice(mi * ncols_per, (mi + 1) * ncols_per)
-                est = sk_clone_without_sc(self)
-                est.set_params(**candidate_params[ci])
-                est._store_fitted(ds, Wh[:, cols], is_clf)
-                est.n_features_in_ = ds.f
-                est.fit_time_ = per
-                refit_local[ci] = est
+            full_ids = sorted(full_models)
+            cols_np = np.asarray([
+                full_models[ci] * ncols_per + j
+                for ci in full_ids for j in range(ncols_per)
+            ])
+            import torch as _torch
+
+            Wfull = W.index_select(
+                1, _torch.as_tensor(cols_np, device=ds.device)
+            ).cpu().numpy()
+            refit_local = {
+                "ids": {ci: k for k, ci in enumerate(full_ids)},
+                "W": Wfull, "per": per,
+            }
         if cluster is not None:
             out = cluster.gather_task_results(results, n_cand * n_folds)
-            boxes = [None] * cluster.world_size
-            if cluster.distributed:
-                import torch.distributed as dist
-
-                dist.all_gather_object(boxes, refit_local)
-                refit_all = {}
-                for b in boxes:
-                    refit_all.update(b)
-            else:
-                refit_all = refit_local
         else:
             out = [results[i] for i in range(n_cand * n_folds)]
-            refit_all = refit_local
-        return {"tasks": out, "refit_estimators": refit_all}
+
+        ncols_per = len(cls)
+        proto = self
+
+        def refit_fn(ci):
+            """Materialize the fitted estimator for candidate ``ci`` from
+            its full-data weight columns (owner rank broadcasts)."""
+            est = None
+            if refit_local and ci in refit_local["ids"]:
+                k2 = refit_local["ids"][ci]
+                cols = slice(k2 * ncols_per, (k2 + 1) * ncols_per)
+                est = sk_clone_without_sc(proto)
+                est.set_params(**candidate_params[ci])
+                est._store_fitted(ds, refit_local["W"][:, cols], is_clf)
+                est.n_features_in_ = ds.f
+                est.fit_time_ = refit_local["per"]
+            if cluster is not None and cluster.distributed:
+                est = cluster.bcast_obj(
+                    est, src=ci % cluster.world_size)
+            return est
+
+        return {"tasks": out, "refit_fn": refit_fn}
 
     def batched_multiclass_fit(self, X, y, cluster, mode="ovr"):
         """Train every one-vs-rest class (or one-vs-one pair) binary
