@@ -71,8 +71,7 @@ extern "C" __global__ __launch_bounds__(256) void k_fwd_gt(
     int loss_id)
 {
     extern __shared__ __attribute__((aligned(16))) char sm[];
-    #define bufA(b) ((__bf16*)(sm + (b) * 16384))
-    #define bufB(b) ((__bf16*)(sm + 8192 + (b) * 16384))
+    #define bufA(b) ((__bf16*)(sm + (b) * 8192))
     __bf16* ldsC = (__bf16*)sm;             // [BN][LDC], aliases the bufs
     char* meta = sm + BN * LDC * 2;
     float* y_s = (float*)meta;              // [BM]
@@ -111,28 +110,38 @@ extern "C" __global__ __launch_bounds__(256) void k_fwd_gt(
     const int nk = fa / BK;
     int cur = 0;
     STAGE_TILE(bufA(0), Abase + (long long)row * fa);
-    STAGE_TILE(bufB(0), Bbase + (long long)row * fa);
+    // B fragments come straight from global (L2-resident W): per-lane
+    // 16-B loads, 4-lane groups cover 64 contiguous bytes per column —
+    // halves the LDS-DMA demand the PMC run showed this kernel parked on
+    bf16x8 bcur[4], bnxt[4];
+    #pragma unroll
+    for (int ni = 0; ni < 4; ++ni)
+        bcur[ni] = *(const bf16x8*)(
+            Bbase + (long long)(wc + ni * 16 + fr) * fa + fk);
     for (int kt = 0; kt < nk; ++kt) {
         __syncthreads();  // staged tile (glds) complete for buf[cur]
         if (kt + 1 < nk) {
             STAGE_TILE(bufA(cur ^ 1),
                        Abase + (long long)row * fa + (kt + 1) * BK);
-            STAGE_TILE(bufB(cur ^ 1),
-                       Bbase + (long long)row * fa + (kt + 1) * BK);
+            #pragma unroll
+            for (int ni = 0; ni < 4; ++ni)
+                bnxt[ni] = *(const bf16x8*)(
+                    Bbase + (long long)(wc + ni * 16 + fr) * fa +
+                    (kt + 1) * BK + fk);
         }
-        bf16x8 af[4], bfr[4];
+        bf16x8 af[4];
         #pragma unroll
         for (int mi = 0; mi < 4; ++mi)
             af[mi] = frag_load(bufA(cur), wr + mi * 16 + fr, fk);
-        #pragma unroll
-        for (int ni = 0; ni < 4; ++ni)
-            bfr[ni] = frag_load(bufB(cur), wc + ni * 16 + fr, fk);
         #pragma unroll
         for (int mi = 0; mi < 4; ++mi)
             #pragma unroll
             for (int ni = 0; ni < 4; ++ni)
                 acc[mi][ni] =
-                    MFMA_BF16_16x16x32(af[mi], bfr[ni], acc[mi][ni]);
+                    MFMA_BF16_16x16x32(af[mi], bcur[ni], acc[mi][ni]);
+        #pragma unroll
+        for (int ni = 0; ni < 4; ++ni)
+            bcur[ni] = bnxt[ni];
         cur ^= 1;
     }
 
@@ -214,31 +223,39 @@ extern "C" __global__ __launch_bounds__(256) void k_grad_partial(
 
     f32x4 acc[4][4] = {};
     int cur = 0;
+    bf16x8 bcur[4], bnxt[4];
     if (k0 < k1) {
         STAGE_TILE(bufA(0), Abase + (long long)row * n_pad + k0);
-        STAGE_TILE(bufB(0), Bbase + (long long)row * gt_stride + k0);
+        #pragma unroll
+        for (int ni = 0; ni < 4; ++ni)
+            bcur[ni] = *(const bf16x8*)(
+                Bbase + (long long)(wc + ni * 16 + fr) * gt_stride + k0 +
+                fk);
     }
     for (int kt = k0; kt < k1; kt += BK) {
         __syncthreads();
         if (kt + BK < k1) {
             STAGE_TILE(bufA(cur ^ 1),
                        Abase + (long long)row * n_pad + kt + BK);
-            STAGE_TILE(bufB(cur ^ 1),
-                       Bbase + (long long)row * gt_stride + kt + BK);
+            #pragma unroll
+            for (int ni = 0; ni < 4; ++ni)
+                bnxt[ni] = *(const bf16x8*)(
+                    Bbase + (long long)(wc + ni * 16 + fr) * gt_stride +
+                    kt + BK + fk);
         }
-        bf16x8 af[4], bfr[4];
+        bf16x8 af[4];
         #pragma unroll
         for (int mi = 0; mi < 4; ++mi)
             af[mi] = frag_load(bufA(cur), wr + mi * 16 + fr, fk);
-        #pragma unroll
-        for (int ni = 0; ni < 4; ++ni)
-            bfr[ni] = frag_load(bufB(cur), wc + ni * 16 + fr, fk);
         #pragma unroll
         for (int mi = 0; mi < 4; ++mi)
             #pragma unroll
             for (int ni = 0; ni < 4; ++ni)
                 acc[mi][ni] =
-                    MFMA_BF16_16x16x32(af[mi], bfr[ni], acc[mi][ni]);
+                    MFMA_BF16_16x16x32(af[mi], bcur[ni], acc[mi][ni]);
+        #pragma unroll
+        for (int ni = 0; ni < 4; ++ni)
+            bcur[ni] = bnxt[ni];
         cur ^= 1;
     }
 

@@ -1,21 +1,10 @@
 export TMPDIR=/tmp
 mkdir -p /root/repo/gpurun_out
 cd /root/repo
-echo "== hash gpu tests =="
-timeout 600 python -m pytest tests/test_hash_gpu.py -x -q > gpurun_out/pytest_hash.log 2>&1; echo "rc=$?"
-tail -6 gpurun_out/pytest_hash.log
-echo "== full gpu suite =="
-timeout 900 python -m pytest tests -m gpu -x -q > gpurun_out/pytest_gpu.log 2>&1; echo "rc=$?"
-tail -2 gpurun_out/pytest_gpu.log
-echo "== bench default =="
-timeout 700 python bench.py --steps 2 --warmup 1 2>/dev/null | tail -1 | tee gpurun_out/bench_b8k.json
-echo "== bench batch 16384 =="
-timeout 700 python bench.py --steps 2 --warmup 1 --batch-size 16384 2>/dev/null | tail -1 | tee gpurun_out/bench_b16k.json
-echo "== bench batch 32768 =="
-timeout 700 python bench.py --steps 2 --warmup 1 --batch-size 32768 2>/dev/null | tail -1 | tee gpurun_out/bench_b32k.json
-echo "== rocprof steady-state (steps 3 warmup 1) =="
+echo "== quick regression: bench after scoring/refit opts =="
+timeout 700 python bench.py --steps 2 --warmup 1 2>/dev/null | tail -1 | tee gpurun_out/bench_opt1.json
+echo "== pmc counters on short bench =="
 cd /tmp
-timeout 900 rocprofv3 --kernel-trace --stats -d /root/repo/gpurun_out/prof2 -o prof2 -- python /root/repo/bench.py --steps 3 --warmup 1 > /root/repo/gpurun_out/prof2.log 2>&1; echo "rocprof rc=$?"
+timeout 900 rocprofv3 --kernel-trace --pmc SQ_WAVE_CYCLES SQ_WAIT_ANY SQ_WAIT_INST_ANY SQ_ACTIVE_INST_ANY SQ_VALU_MFMA_BUSY_CYCLES -d /root/repo/gpurun_out/pmc1 -o pmc1 -- python /root/repo/bench.py --steps 1 --warmup 1 --candidates-per-gpu 500 --epochs 4 > /root/repo/gpurun_out/pmc1.log 2>&1; echo "pmc rc=$?"
 cd /root/repo
-python tools/prof_summary.py gpurun_out/prof2/prof2_results.db > gpurun_out/prof2_summary.txt 2>&1
-head -25 gpurun_out/prof2_summary.txt
+python tools/pmc_summary.py gpurun_out/pmc1 2>&1 | head -40 | tee gpurun_out/pmc1_summary.txt
