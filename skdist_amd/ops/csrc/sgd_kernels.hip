@@ -54,6 +54,20 @@ frag_load(const __bf16* lds, int row, int fk) {
     return *(const bf16x8*)(lds + row * BK + phys * 8);
 }
 
+// Software-pipelined sync: wait only the tiles this iteration consumes
+// (hipcc's __syncthreads() drains vmcnt(0) whenever LDS-DMA is in
+// flight, which serializes every prefetch — PMC showed both GEMMs ~50%
+// parked on exactly that).  vmcnt(N) keeps the N newest VMEM ops (the
+// prefetched A-glds + B fragment loads) in flight across the barrier;
+// per-wave issue counts are uniform so the raw s_barrier is safe.
+#define SYNC_KEEP(n)                                                       \
+    do {                                                                   \
+        asm volatile("s_waitcnt vmcnt(" #n ")" ::: "memory");              \
+        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");                 \
+        asm volatile("s_barrier" ::: "memory");                            \
+    } while (0)
+#define ISSUE_FENCE() asm volatile("" ::: "memory")
+
 // ---------------------------------------------------------------------- //
 // K1: fused forward GEMM + loss gradient + transposed store
 // grid: (m_pad/BM, ncols_pad/BN), block 256
@@ -71,7 +85,7 @@ extern "C" __global__ __launch_bounds__(256) void k_fwd_gt(
     int loss_id)
 {
     extern __shared__ __attribute__((aligned(16))) char sm[];
-    #define bufA(b) ((__bf16*)(sm + (b) * 8192))
+    #define bufA(b) ((__bf16*)(sm + (b) * 8192))   // 4-deep A ring
     __bf16* ldsC = (__bf16*)sm;             // [BN][LDC], aliases the bufs
     char* meta = sm + BN * LDC * 2;
     float* y_s = (float*)meta;              // [BM]
@@ -108,31 +122,43 @@ extern "C" __global__ __launch_bounds__(256) void k_fwd_gt(
 
     f32x4 acc[4][4] = {};
     const int nk = fa / BK;
-    int cur = 0;
+    // pipeline: A staged 3 tiles ahead through a 4-deep LDS ring; B (the
+    // L2-resident W panel) read as per-lane 16-B fragment loads one tile
+    // ahead.  Prologue issue order: A(0), B(0), A(1), A(2).
+    const long long bcol = (long long)(wc + fr) * fa;
     STAGE_TILE(bufA(0), Abase + (long long)row * fa);
-    // B fragments come straight from global (L2-resident W): per-lane
-    // 16-B loads, 4-lane groups cover 64 contiguous bytes per column —
-    // halves the LDS-DMA demand the PMC run showed this kernel parked on
+    ISSUE_FENCE();
     bf16x8 bcur[4], bnxt[4];
     #pragma unroll
     for (int ni = 0; ni < 4; ++ni)
-        bcur[ni] = *(const bf16x8*)(
-            Bbase + (long long)(wc + ni * 16 + fr) * fa + fk);
+        bcur[ni] = *(const bf16x8*)(Bbase + bcol + (long long)ni * 16 * fa
+                                    + fk);
+    ISSUE_FENCE();
+    if (1 < nk) STAGE_TILE(bufA(1), Abase + (long long)row * fa + BK);
+    ISSUE_FENCE();
+    if (2 < nk)
+        STAGE_TILE(bufA(2), Abase + (long long)row * fa + 2 * BK);
     for (int kt = 0; kt < nk; ++kt) {
-        __syncthreads();  // staged tile (glds) complete for buf[cur]
         if (kt + 1 < nk) {
-            STAGE_TILE(bufA(cur ^ 1),
-                       Abase + (long long)row * fa + (kt + 1) * BK);
             #pragma unroll
             for (int ni = 0; ni < 4; ++ni)
                 bnxt[ni] = *(const bf16x8*)(
-                    Bbase + (long long)(wc + ni * 16 + fr) * fa +
+                    Bbase + bcol + (long long)ni * 16 * fa +
                     (kt + 1) * BK + fk);
+        }
+        ISSUE_FENCE();
+        if (kt + 3 < nk)
+            STAGE_TILE(bufA((kt + 3) & 3),
+                       Abase + (long long)row * fa + (kt + 3) * BK);
+        if (kt + 3 < nk) {
+            SYNC_KEEP(8);   // keep {A(kt+2), B(kt+1), A(kt+3)} in flight
+        } else {
+            SYNC_KEEP(0);   // pipeline tail: full drain
         }
         bf16x8 af[4];
         #pragma unroll
         for (int mi = 0; mi < 4; ++mi)
-            af[mi] = frag_load(bufA(cur), wr + mi * 16 + fr, fk);
+            af[mi] = frag_load(bufA(kt & 3), wr + mi * 16 + fr, fk);
         #pragma unroll
         for (int mi = 0; mi < 4; ++mi)
             #pragma unroll
@@ -142,7 +168,6 @@ extern "C" __global__ __launch_bounds__(256) void k_fwd_gt(
         #pragma unroll
         for (int ni = 0; ni < 4; ++ni)
             bcur[ni] = bnxt[ni];
-        cur ^= 1;
     }
 
     // epilogue: z -> masked dloss -> bf16, transposed through LDS
@@ -222,31 +247,46 @@ extern "C" __global__ __launch_bounds__(256) void k_grad_partial(
     const int fk = (lane >> 4) * 8;
 
     f32x4 acc[4][4] = {};
-    int cur = 0;
+    const int nk = (k1 - k0) / BK;   // k_chunk is BK-aligned
+    const long long bcol = (long long)(wc + fr) * gt_stride;
     bf16x8 bcur[4], bnxt[4];
-    if (k0 < k1) {
+    if (nk > 0) {
         STAGE_TILE(bufA(0), Abase + (long long)row * n_pad + k0);
+        ISSUE_FENCE();
         #pragma unroll
         for (int ni = 0; ni < 4; ++ni)
             bcur[ni] = *(const bf16x8*)(
-                Bbase + (long long)(wc + ni * 16 + fr) * gt_stride + k0 +
-                fk);
+                Bbase + bcol + (long long)ni * 16 * gt_stride + k0 + fk);
+        ISSUE_FENCE();
+        if (1 < nk)
+            STAGE_TILE(bufA(1), Abase + (long long)row * n_pad + k0 + BK);
+        ISSUE_FENCE();
+        if (2 < nk)
+            STAGE_TILE(bufA(2),
+                       Abase + (long long)row * n_pad + k0 + 2 * BK);
     }
-    for (int kt = k0; kt < k1; kt += BK) {
-        __syncthreads();
-        if (kt + BK < k1) {
-            STAGE_TILE(bufA(cur ^ 1),
-                       Abase + (long long)row * n_pad + kt + BK);
+    for (int it = 0; it < nk; ++it) {
+        const int kt = k0 + it * BK;
+        if (it + 1 < nk) {
             #pragma unroll
             for (int ni = 0; ni < 4; ++ni)
                 bnxt[ni] = *(const bf16x8*)(
-                    Bbase + (long long)(wc + ni * 16 + fr) * gt_stride +
-                    kt + BK + fk);
+                    Bbase + bcol + (long long)ni * 16 * gt_stride + kt +
+                    BK + fk);
+        }
+        ISSUE_FENCE();
+        if (it + 3 < nk)
+            STAGE_TILE(bufA((it + 3) & 3),
+                       Abase + (long long)row * n_pad + kt + 3 * BK);
+        if (it + 3 < nk) {
+            SYNC_KEEP(8);
+        } else {
+            SYNC_KEEP(0);
         }
         bf16x8 af[4];
         #pragma unroll
         for (int mi = 0; mi < 4; ++mi)
-            af[mi] = frag_load(bufA(cur), wr + mi * 16 + fr, fk);
+            af[mi] = frag_load(bufA(it & 3), wr + mi * 16 + fr, fk);
         #pragma unroll
         for (int mi = 0; mi < 4; ++mi)
             #pragma unroll
@@ -256,7 +296,6 @@ extern "C" __global__ __launch_bounds__(256) void k_grad_partial(
         #pragma unroll
         for (int ni = 0; ni < 4; ++ni)
             bcur[ni] = bnxt[ni];
-        cur ^= 1;
     }
 
     float* out = partial + (long long)z * fa * ncols_pad;
