@@ -85,9 +85,10 @@ extern "C" __global__ __launch_bounds__(256) void k_fwd_gt(
     int loss_id)
 {
     extern __shared__ __attribute__((aligned(16))) char sm[];
-    #define bufA(b) ((__bf16*)(sm + (b) * 8192))   // 4-deep A ring
+    #define bufA(b) ((__bf16*)(sm + (b) * 8192))           // 3-deep rings
+    #define bufB(b) ((__bf16*)(sm + 24576 + (b) * 8192))
     __bf16* ldsC = (__bf16*)sm;             // [BN][LDC], aliases the bufs
-    char* meta = sm + BN * LDC * 2;
+    char* meta = sm + 49152;
     float* y_s = (float*)meta;              // [BM]
     int* fold_s = (int*)(meta + 512);       // [BM]
     int* cls_s = (int*)(meta + 1024);       // [BN]
@@ -122,52 +123,47 @@ extern "C" __global__ __launch_bounds__(256) void k_fwd_gt(
 
     f32x4 acc[4][4] = {};
     const int nk = fa / BK;
-    // pipeline: A staged 3 tiles ahead through a 4-deep LDS ring; B (the
-    // L2-resident W panel) read as per-lane 16-B fragment loads one tile
-    // ahead.  Prologue issue order: A(0), B(0), A(1), A(2).
-    const long long bcol = (long long)(wc + fr) * fa;
+    // pipeline: both operands staged 2 tiles ahead through 3-deep LDS
+    // rings; SYNC_KEEP waits only the pair being consumed, so the two
+    // prefetched pairs stay in flight across the barrier.
     STAGE_TILE(bufA(0), Abase + (long long)row * fa);
+    STAGE_TILE(bufB(0), Bbase + (long long)row * fa);
     ISSUE_FENCE();
-    bf16x8 bcur[4], bnxt[4];
-    #pragma unroll
-    for (int ni = 0; ni < 4; ++ni)
-        bcur[ni] = *(const bf16x8*)(Bbase + bcol + (long long)ni * 16 * fa
-                                    + fk);
-    ISSUE_FENCE();
-    if (1 < nk) STAGE_TILE(bufA(1), Abase + (long long)row * fa + BK);
-    ISSUE_FENCE();
-    if (2 < nk)
-        STAGE_TILE(bufA(2), Abase + (long long)row * fa + 2 * BK);
+    if (1 < nk) {
+        STAGE_TILE(bufA(1), Abase + (long long)row * fa + BK);
+        STAGE_TILE(bufB(1), Bbase + (long long)row * fa + BK);
+    }
     for (int kt = 0; kt < nk; ++kt) {
+        // barrier first: every wave is done reading ring slot (kt-1)%3
+        // before it gets restaged below; wait covers only pair (kt)
         if (kt + 1 < nk) {
-            #pragma unroll
-            for (int ni = 0; ni < 4; ++ni)
-                bnxt[ni] = *(const bf16x8*)(
-                    Bbase + bcol + (long long)ni * 16 * fa +
-                    (kt + 1) * BK + fk);
+            SYNC_KEEP(4);   // pair (kt+1) stays in flight
+        } else {
+            SYNC_KEEP(0);
         }
         ISSUE_FENCE();
-        if (kt + 3 < nk)
-            STAGE_TILE(bufA((kt + 3) & 3),
-                       Abase + (long long)row * fa + (kt + 3) * BK);
-        if (kt + 3 < nk) {
-            SYNC_KEEP(8);   // keep {A(kt+2), B(kt+1), A(kt+3)} in flight
-        } else {
-            SYNC_KEEP(0);   // pipeline tail: full drain
+        if (kt + 2 < nk) {
+            const int b = (kt + 2) % 3;
+            STAGE_TILE(bufA(b), Abase + (long long)row * fa +
+                                    (kt + 2) * BK);
+            STAGE_TILE(bufB(b), Bbase + (long long)row * fa +
+                                    (kt + 2) * BK);
         }
-        bf16x8 af[4];
+        ISSUE_FENCE();
+        const int cur = kt % 3;
+        bf16x8 af[4], bfr[4];
         #pragma unroll
         for (int mi = 0; mi < 4; ++mi)
-            af[mi] = frag_load(bufA(kt & 3), wr + mi * 16 + fr, fk);
+            af[mi] = frag_load(bufA(cur), wr + mi * 16 + fr, fk);
+        #pragma unroll
+        for (int ni = 0; ni < 4; ++ni)
+            bfr[ni] = frag_load(bufB(cur), wc + ni * 16 + fr, fk);
         #pragma unroll
         for (int mi = 0; mi < 4; ++mi)
             #pragma unroll
             for (int ni = 0; ni < 4; ++ni)
                 acc[mi][ni] =
-                    MFMA_BF16_16x16x32(af[mi], bcur[ni], acc[mi][ni]);
-        #pragma unroll
-        for (int ni = 0; ni < 4; ++ni)
-            bcur[ni] = bnxt[ni];
+                    MFMA_BF16_16x16x32(af[mi], bfr[ni], acc[mi][ni]);
     }
 
     // epilogue: z -> masked dloss -> bf16, transposed through LDS
@@ -248,54 +244,47 @@ extern "C" __global__ __launch_bounds__(256) void k_grad_partial(
 
     f32x4 acc[4][4] = {};
     const int nk = (k1 - k0) / BK;   // k_chunk is BK-aligned
-    const long long bcol = (long long)(wc + fr) * gt_stride;
-    bf16x8 bcur[4], bnxt[4];
     if (nk > 0) {
         STAGE_TILE(bufA(0), Abase + (long long)row * n_pad + k0);
+        STAGE_TILE(bufB(0), Bbase + (long long)row * gt_stride + k0);
         ISSUE_FENCE();
-        #pragma unroll
-        for (int ni = 0; ni < 4; ++ni)
-            bcur[ni] = *(const bf16x8*)(
-                Bbase + bcol + (long long)ni * 16 * gt_stride + k0 + fk);
-        ISSUE_FENCE();
-        if (1 < nk)
-            STAGE_TILE(bufA(1), Abase + (long long)row * n_pad + k0 + BK);
-        ISSUE_FENCE();
-        if (2 < nk)
-            STAGE_TILE(bufA(2),
-                       Abase + (long long)row * n_pad + k0 + 2 * BK);
+        if (1 < nk) {
+            STAGE_TILE(bufA(1),
+                       Abase + (long long)row * n_pad + k0 + BK);
+            STAGE_TILE(bufB(1),
+                       Bbase + (long long)row * gt_stride + k0 + BK);
+        }
     }
     for (int it = 0; it < nk; ++it) {
         const int kt = k0 + it * BK;
         if (it + 1 < nk) {
-            #pragma unroll
-            for (int ni = 0; ni < 4; ++ni)
-                bnxt[ni] = *(const bf16x8*)(
-                    Bbase + bcol + (long long)ni * 16 * gt_stride + kt +
-                    BK + fk);
-        }
-        ISSUE_FENCE();
-        if (it + 3 < nk)
-            STAGE_TILE(bufA((it + 3) & 3),
-                       Abase + (long long)row * n_pad + kt + 3 * BK);
-        if (it + 3 < nk) {
-            SYNC_KEEP(8);
+            SYNC_KEEP(4);
         } else {
             SYNC_KEEP(0);
         }
-        bf16x8 af[4];
+        ISSUE_FENCE();
+        if (it + 2 < nk) {
+            const int b = (it + 2) % 3;
+            STAGE_TILE(bufA(b),
+                       Abase + (long long)row * n_pad + kt + 2 * BK);
+            STAGE_TILE(bufB(b),
+                       Bbase + (long long)row * gt_stride + kt + 2 * BK);
+        }
+        ISSUE_FENCE();
+        const int cur = it % 3;
+        bf16x8 af[4], bfr[4];
         #pragma unroll
         for (int mi = 0; mi < 4; ++mi)
-            af[mi] = frag_load(bufA(it & 3), wr + mi * 16 + fr, fk);
+            af[mi] = frag_load(bufA(cur), wr + mi * 16 + fr, fk);
+        #pragma unroll
+        for (int ni = 0; ni < 4; ++ni)
+            bfr[ni] = frag_load(bufB(cur), wc + ni * 16 + fr, fk);
         #pragma unroll
         for (int mi = 0; mi < 4; ++mi)
             #pragma unroll
             for (int ni = 0; ni < 4; ++ni)
                 acc[mi][ni] =
-                    MFMA_BF16_16x16x32(af[mi], bcur[ni], acc[mi][ni]);
-        #pragma unroll
-        for (int ni = 0; ni < 4; ++ni)
-            bcur[ni] = bnxt[ni];
+                    MFMA_BF16_16x16x32(af[mi], bfr[ni], acc[mi][ni]);
     }
 
     float* out = partial + (long long)z * fa * ncols_pad;
@@ -370,7 +359,7 @@ extern "C" hipError_t skdist_sgd_step(
     const int m_pad = (int)((m + BM - 1) / BM) * BM;
     {
         dim3 grid(m_pad / BM, ncols_pad / BN);
-        size_t lds = (size_t)BN * LDC * 2 + 2560;
+        size_t lds = 49152 + 2560;  // 6 ring bufs (ldsC aliases) + meta
         hipLaunchKernelGGL(k_fwd_gt, grid, dim3(256), lds, stream,
                            (const __bf16*)Xs, (const __bf16*)WbfT_in,
                            (__bf16*)GT, (const float*)y, (const int*)fold,
@@ -383,7 +372,7 @@ extern "C" hipError_t skdist_sgd_step(
     {
         const int k_chunk = (int)((m_pad / splitk + BK - 1) / BK) * BK;
         dim3 grid((unsigned)(fa_store / BM), ncols_pad / BN, splitk);
-        size_t lds = 32768;
+        size_t lds = 49152;  // 2 x 3-deep 8 KiB rings
         hipLaunchKernelGGL(k_grad_partial, grid, dim3(256), lds, stream,
                            (const __bf16*)XsT, (const __bf16*)GT,
                            (float*)partial, (int)start, m_pad, n_pad, fa,
