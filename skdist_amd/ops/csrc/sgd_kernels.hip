@@ -54,20 +54,6 @@ frag_load(const __bf16* lds, int row, int fk) {
     return *(const bf16x8*)(lds + row * BK + phys * 8);
 }
 
-// Software-pipelined sync: wait only the tiles this iteration consumes
-// (hipcc's __syncthreads() drains vmcnt(0) whenever LDS-DMA is in
-// flight, which serializes every prefetch — PMC showed both GEMMs ~50%
-// parked on exactly that).  vmcnt(N) keeps the N newest VMEM ops (the
-// prefetched A-glds + B fragment loads) in flight across the barrier;
-// per-wave issue counts are uniform so the raw s_barrier is safe.
-#define SYNC_KEEP(n)                                                       \
-    do {                                                                   \
-        asm volatile("s_waitcnt vmcnt(" #n ")" ::: "memory");              \
-        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");                 \
-        asm volatile("s_barrier" ::: "memory");                            \
-    } while (0)
-#define ISSUE_FENCE() asm volatile("" ::: "memory")
-
 // ---------------------------------------------------------------------- //
 // K1: fused forward GEMM + loss gradient + transposed store
 // grid: (m_pad/BM, ncols_pad/BN), block 256
@@ -85,10 +71,10 @@ extern "C" __global__ __launch_bounds__(256) void k_fwd_gt(
     int loss_id)
 {
     extern __shared__ __attribute__((aligned(16))) char sm[];
-    #define bufA(b) ((__bf16*)(sm + (b) * 8192))           // 3-deep rings
-    #define bufB(b) ((__bf16*)(sm + 24576 + (b) * 8192))
+    #define bufA(b) ((__bf16*)(sm + (b) * 16384))
+    #define bufB(b) ((__bf16*)(sm + 8192 + (b) * 16384))
     __bf16* ldsC = (__bf16*)sm;             // [BN][LDC], aliases the bufs
-    char* meta = sm + 49152;
+    char* meta = sm + BN * LDC * 2;
     float* y_s = (float*)meta;              // [BM]
     int* fold_s = (int*)(meta + 512);       // [BM]
     int* cls_s = (int*)(meta + 1024);       // [BN]
@@ -123,34 +109,17 @@ extern "C" __global__ __launch_bounds__(256) void k_fwd_gt(
 
     f32x4 acc[4][4] = {};
     const int nk = fa / BK;
-    // pipeline: both operands staged 2 tiles ahead through 3-deep LDS
-    // rings; SYNC_KEEP waits only the pair being consumed, so the two
-    // prefetched pairs stay in flight across the barrier.
+    int cur = 0;
     STAGE_TILE(bufA(0), Abase + (long long)row * fa);
     STAGE_TILE(bufB(0), Bbase + (long long)row * fa);
-    ISSUE_FENCE();
-    if (1 < nk) {
-        STAGE_TILE(bufA(1), Abase + (long long)row * fa + BK);
-        STAGE_TILE(bufB(1), Bbase + (long long)row * fa + BK);
-    }
     for (int kt = 0; kt < nk; ++kt) {
-        // barrier first: every wave is done reading ring slot (kt-1)%3
-        // before it gets restaged below; wait covers only pair (kt)
+        __syncthreads();  // staged tile (glds) complete for buf[cur]
         if (kt + 1 < nk) {
-            SYNC_KEEP(4);   // pair (kt+1) stays in flight
-        } else {
-            SYNC_KEEP(0);
+            STAGE_TILE(bufA(cur ^ 1),
+                       Abase + (long long)row * fa + (kt + 1) * BK);
+            STAGE_TILE(bufB(cur ^ 1),
+                       Bbase + (long long)row * fa + (kt + 1) * BK);
         }
-        ISSUE_FENCE();
-        if (kt + 2 < nk) {
-            const int b = (kt + 2) % 3;
-            STAGE_TILE(bufA(b), Abase + (long long)row * fa +
-                                    (kt + 2) * BK);
-            STAGE_TILE(bufB(b), Bbase + (long long)row * fa +
-                                    (kt + 2) * BK);
-        }
-        ISSUE_FENCE();
-        const int cur = kt % 3;
         bf16x8 af[4], bfr[4];
         #pragma unroll
         for (int mi = 0; mi < 4; ++mi)
@@ -164,6 +133,7 @@ extern "C" __global__ __launch_bounds__(256) void k_fwd_gt(
             for (int ni = 0; ni < 4; ++ni)
                 acc[mi][ni] =
                     MFMA_BF16_16x16x32(af[mi], bfr[ni], acc[mi][ni]);
+        cur ^= 1;
     }
 
     // epilogue: z -> masked dloss -> bf16, transposed through LDS
@@ -243,35 +213,19 @@ extern "C" __global__ __launch_bounds__(256) void k_grad_partial(
     const int fk = (lane >> 4) * 8;
 
     f32x4 acc[4][4] = {};
-    const int nk = (k1 - k0) / BK;   // k_chunk is BK-aligned
-    if (nk > 0) {
+    int cur = 0;
+    if (k0 < k1) {
         STAGE_TILE(bufA(0), Abase + (long long)row * n_pad + k0);
         STAGE_TILE(bufB(0), Bbase + (long long)row * gt_stride + k0);
-        ISSUE_FENCE();
-        if (1 < nk) {
-            STAGE_TILE(bufA(1),
-                       Abase + (long long)row * n_pad + k0 + BK);
-            STAGE_TILE(bufB(1),
-                       Bbase + (long long)row * gt_stride + k0 + BK);
-        }
     }
-    for (int it = 0; it < nk; ++it) {
-        const int kt = k0 + it * BK;
-        if (it + 1 < nk) {
-            SYNC_KEEP(4);
-        } else {
-            SYNC_KEEP(0);
+    for (int kt = k0; kt < k1; kt += BK) {
+        __syncthreads();
+        if (kt + BK < k1) {
+            STAGE_TILE(bufA(cur ^ 1),
+                       Abase + (long long)row * n_pad + kt + BK);
+            STAGE_TILE(bufB(cur ^ 1),
+                       Bbase + (long long)row * gt_stride + kt + BK);
         }
-        ISSUE_FENCE();
-        if (it + 2 < nk) {
-            const int b = (it + 2) % 3;
-            STAGE_TILE(bufA(b),
-                       Abase + (long long)row * n_pad + kt + 2 * BK);
-            STAGE_TILE(bufB(b),
-                       Bbase + (long long)row * gt_stride + kt + 2 * BK);
-        }
-        ISSUE_FENCE();
-        const int cur = it % 3;
         bf16x8 af[4], bfr[4];
         #pragma unroll
         for (int mi = 0; mi < 4; ++mi)
@@ -285,6 +239,7 @@ extern "C" __global__ __launch_bounds__(256) void k_grad_partial(
             for (int ni = 0; ni < 4; ++ni)
                 acc[mi][ni] =
                     MFMA_BF16_16x16x32(af[mi], bfr[ni], acc[mi][ni]);
+        cur ^= 1;
     }
 
     float* out = partial + (long long)z * fa * ncols_pad;
@@ -359,7 +314,7 @@ extern "C" hipError_t skdist_sgd_step(
     const int m_pad = (int)((m + BM - 1) / BM) * BM;
     {
         dim3 grid(m_pad / BM, ncols_pad / BN);
-        size_t lds = 49152 + 2560;  // 6 ring bufs (ldsC aliases) + meta
+        size_t lds = (size_t)BN * LDC * 2 + 2560;
         hipLaunchKernelGGL(k_fwd_gt, grid, dim3(256), lds, stream,
                            (const __bf16*)Xs, (const __bf16*)WbfT_in,
                            (__bf16*)GT, (const float*)y, (const int*)fold,
@@ -372,7 +327,7 @@ extern "C" hipError_t skdist_sgd_step(
     {
         const int k_chunk = (int)((m_pad / splitk + BK - 1) / BK) * BK;
         dim3 grid((unsigned)(fa_store / BM), ncols_pad / BN, splitk);
-        size_t lds = 49152;  // 2 x 3-deep 8 KiB rings
+        size_t lds = 32768;
         hipLaunchKernelGGL(k_grad_partial, grid, dim3(256), lds, stream,
                            (const __bf16*)XsT, (const __bf16*)GT,
                            (float*)partial, (int)start, m_pad, n_pad, fa,
