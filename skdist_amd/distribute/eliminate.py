@@ -97,36 +97,58 @@ class DistFeatureEliminator(ClassifierMixin, BaseEstimator):
             removals.append(ranks[:removed])
 
         cv_splits = list(cv.split(X, y, groups))
-        tasks = [
-            (tid, si, split)
-            for tid, (si, split) in enumerate(
-                product(range(len(removals)), cv_splits)
-            )
-        ]
 
-        def task_fn(task):
-            tid, si, (train, test) = task
-            est = _clone(self.estimator)
-            if hasattr(est, "sc"):
-                est.sc = None
-            Xs = _drop_cols(X, removals[si])
-            X_tr, y_tr = _safe_split(est, Xs, y, train)
-            X_te, y_te = _safe_split(est, Xs, y, test, train)
-            est.fit(X_tr, y_tr, **fit_params)
-            return tid, si, float(scorer(est, X_te, y_te))
+        # batched device path: ONE masked multi-column solve scores every
+        # (subset × fold) — replaces the reference's per-subset Spark
+        # tasks (eliminate.py:191-210)
+        refit_fn = None
+        scores_mat = None
+        if (
+            sc is not None
+            and hasattr(self.estimator, "batched_eliminate")
+            and not fit_params
+        ):
+            from ..models.linear import FallbackToGeneric
 
-        if sc is None:
-            results = run_local_tasks(
-                task_fn, tasks, n_jobs=self.n_jobs,
-                pre_dispatch=self.pre_dispatch or "2*n_jobs",
-            )
+            try:
+                scores_mat, refit_fn = self.estimator.batched_eliminate(
+                    X, y, removals, cv_splits, self.scoring, sc)
+            except FallbackToGeneric:
+                scores_mat = None
+
+        if scores_mat is not None:
+            self.scores_ = [float(m) for m in scores_mat.mean(axis=1)]
         else:
-            results = sc.run_tasks(task_fn, tasks)
+            tasks = [
+                (tid, si, split)
+                for tid, (si, split) in enumerate(
+                    product(range(len(removals)), cv_splits)
+                )
+            ]
 
-        per_set = [[] for _ in removals]
-        for tid, si, score in sorted(results, key=lambda r: r[0]):
-            per_set[si].append(score)
-        self.scores_ = [float(np.mean(s)) for s in per_set]
+            def task_fn(task):
+                tid, si, (train, test) = task
+                est = _clone(self.estimator)
+                if hasattr(est, "sc"):
+                    est.sc = None
+                Xs = _drop_cols(X, removals[si])
+                X_tr, y_tr = _safe_split(est, Xs, y, train)
+                X_te, y_te = _safe_split(est, Xs, y, test, train)
+                est.fit(X_tr, y_tr, **fit_params)
+                return tid, si, float(scorer(est, X_te, y_te))
+
+            if sc is None:
+                results = run_local_tasks(
+                    task_fn, tasks, n_jobs=self.n_jobs,
+                    pre_dispatch=self.pre_dispatch or "2*n_jobs",
+                )
+            else:
+                results = sc.run_tasks(task_fn, tasks)
+
+            per_set = [[] for _ in removals]
+            for tid, si, score in sorted(results, key=lambda r: r[0]):
+                per_set[si].append(score)
+            self.scores_ = [float(np.mean(s)) for s in per_set]
 
         # exact ties break toward the SMALLEST feature set (the ladder is
         # ordered by increasing removal) — parsimony over argmax-first
@@ -139,10 +161,17 @@ class DistFeatureEliminator(ClassifierMixin, BaseEstimator):
             )
         else:
             self.best_features_ = np.arange(n_features)
-        self.best_estimator_ = _clone(self.estimator)
-        if hasattr(self.best_estimator_, "sc"):
-            self.best_estimator_.sc = None
-        self.best_estimator_.fit(X[:, self.best_features_], y, **fit_params)
+        best_est = refit_fn(best) if refit_fn is not None else None
+        if best_est is not None:
+            # batched path: the subset's full-data model trained alongside
+            # the CV columns — refit is a column extraction
+            self.best_estimator_ = best_est
+        else:
+            self.best_estimator_ = _clone(self.estimator)
+            if hasattr(self.best_estimator_, "sc"):
+                self.best_estimator_.sc = None
+            self.best_estimator_.fit(
+                X[:, self.best_features_], y, **fit_params)
         self.n_features_ = len(self.best_features_)
 
         del self.sc

@@ -208,10 +208,15 @@ class ColumnSpec:
 
     ``col_class2[c] >= 0`` marks a one-vs-one column: only rows with
     ``y in (col_class[c], col_class2[c])`` train it.
+
+    ``feat_mask`` (optional, [fa, ncols] uint8): per-column feature
+    subset — masked rows of W are pinned to 0 after every update, so
+    column c trains exactly the model on its feature subset (the batched
+    DistFeatureEliminator path; intercept/pad rows must be 1).
     """
 
     def __init__(self, device, col_fold, col_class, col_lr, col_l2,
-                 col_class2=None):
+                 col_class2=None, feat_mask=None):
         as_t = lambda a, dt: torch.as_tensor(
             np.ascontiguousarray(a), dtype=dt, device=device
         )
@@ -222,6 +227,9 @@ class ColumnSpec:
         if col_class2 is None:
             col_class2 = np.full(len(col_fold), -1, dtype=np.int32)
         self.col_class2 = as_t(col_class2, torch.int32)
+        self.feat_mask = (
+            None if feat_mask is None else as_t(feat_mask, torch.uint8)
+        )
         self.ncols = len(col_fold)
 
 
@@ -249,6 +257,10 @@ def batched_sgd_fit(ds, spec, loss, epochs, batch_size, seed=0,
     perm = torch.as_tensor(
         rng.permutation(n), dtype=torch.int64, device=device
     )
+    fmask = (
+        spec.feat_mask.to(torch.float32)
+        if spec.feat_mask is not None else None
+    )
     for epoch in range(epochs):
         lr_scale = 1.0 / (1.0 + lr_decay * epoch)
         for start in range(0, n, batch_size):
@@ -257,6 +269,10 @@ def batched_sgd_fit(ds, spec, loss, epochs, batch_size, seed=0,
                 ds.Xaug, ds.y_float, ds.fold_id, idx, W, V, spec,
                 loss_id, lr_scale, momentum, ds.intercept_row,
             )
+            if fmask is not None:
+                W.mul_(fmask)
+                if V is not None:
+                    V.mul_(fmask)
     return W
 
 

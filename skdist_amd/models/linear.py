@@ -296,6 +296,146 @@ class _BatchedLinearBase(BaseEstimator):
 
         return {"tasks": out, "refit_fn": refit_fn}
 
+    def batched_eliminate(self, X, y, removals, cv_splits, scoring,
+                          cluster):
+        """Score every (feature-subset × fold) in ONE masked batched
+        solve (DistFeatureEliminator device path; reference scores each
+        subset as its own Spark task, eliminate.py:191-210).
+
+        Per-column feature masks pin the removed features' weights to 0
+        inside the solver, so column (s, fold) trains exactly the model
+        on subset s.  Returns (scores [n_sets, n_folds], refit_fn) where
+        ``refit_fn(s)`` materializes the fitted estimator for subset s
+        from its full-data column, coef_ sliced to the kept features.
+        """
+        import scipy.sparse as _sp
+
+        if _sp.issparse(X):
+            raise FallbackToGeneric("sparse X not batched")
+        metric = self._device_metric(scoring)
+        is_clf = isinstance(self, ClassifierMixin)
+        t0 = time.perf_counter()
+        ds = DeviceDataset(
+            None if X is None else np.asarray(X, dtype=np.float32),
+            None if y is None else np.asarray(y),
+            cluster=cluster, standardize=self.standardize,
+        )
+        if not ds.set_cv_partition(cv_splits):
+            raise FallbackToGeneric("cv splits do not partition the data")
+        n_folds = len(cv_splits)
+        n_sets = len(removals)
+        if is_clf:
+            k = len(ds.classes_)
+            n_classes = k
+            cls = (
+                np.array([1], dtype=np.int32) if k == 2
+                else np.arange(k, dtype=np.int32)
+            )
+        else:
+            n_classes = 2
+            cls = np.array([-1], dtype=np.int32)
+        cpm = len(cls)
+
+        set_ids = (
+            cluster.shard_indices(n_sets) if cluster is not None
+            else list(range(n_sets))
+        )
+        fold_train_n = [len(tr) for tr, _ in cv_splits]
+        col_fold, col_class, col_lr, col_l2 = [], [], [], []
+        model_folds = []
+        masks = []          # per MODEL (shared by its cpm columns)
+        local_rows = []     # (set_id, model_idx, fold | -2)
+        model_idx = 0
+        for si in set_ids:
+            drop = np.asarray(removals[si], dtype=np.int64)
+            m = np.ones(ds.fa, dtype=np.uint8)
+            m[drop] = 0      # feature j is W row j; intercept/pad stay 1
+            for fold in list(range(n_folds)) + [-2]:
+                n_train = (
+                    fold_train_n[fold] if fold >= 0 else ds.n
+                )
+                lam = self._lam(n_train)
+                for cc in cls:
+                    col_fold.append(fold)
+                    col_class.append(cc)
+                    col_lr.append(self.lr)
+                    col_l2.append(lam)
+                masks.append(m)
+                model_folds.append(fold)
+                local_rows.append((si, model_idx, fold))
+                model_idx += 1
+
+        local_scores = {}
+        refit_local = {}
+        if model_idx:
+            feat_mask = np.repeat(
+                np.stack(masks, axis=1), cpm, axis=1
+            )  # [fa, n_models*cpm]
+            spec = ColumnSpec(
+                ds.device,
+                col_fold=np.asarray(col_fold, dtype=np.int32),
+                col_class=np.asarray(col_class, dtype=np.int32),
+                col_lr=np.asarray(col_lr, dtype=np.float32),
+                col_l2=np.asarray(col_l2, dtype=np.float32),
+                feat_mask=feat_mask,
+            )
+            W = batched_sgd_fit(
+                ds, spec, self._loss, self.epochs, self.batch_size,
+                seed=self._seed(), momentum=self.momentum,
+            )
+            scores = batched_scores_by_fold(
+                ds, W, np.asarray(model_folds),
+                np.asarray(col_class, dtype=np.int32),
+                n_classes=n_classes, metric=metric,
+            )
+            per = (time.perf_counter() - t0) / max(model_idx, 1)
+            Wh = W.cpu().numpy()
+            for si, mi, fold in local_rows:
+                if fold >= 0:
+                    local_scores.setdefault(si, {})[fold] = float(
+                        scores[mi])
+                else:
+                    refit_local[si] = (
+                        Wh[:, mi * cpm:(mi + 1) * cpm], per)
+
+        # all-gather the per-set fold scores
+        if cluster is not None and cluster.distributed:
+            import torch.distributed as dist
+
+            boxes = [None] * cluster.world_size
+            dist.all_gather_object(boxes, local_scores)
+            merged = {}
+            for b in boxes:
+                merged.update(b)
+        else:
+            merged = local_scores
+        out = np.zeros((n_sets, n_folds))
+        for si, folds in merged.items():
+            for fold, sc_v in folds.items():
+                out[si, fold] = sc_v
+
+        proto = self
+
+        def refit_fn(si):
+            est = None
+            if si in refit_local:
+                Wcols, per = refit_local[si]
+                est = sk_clone_without_sc(proto)
+                est._store_fitted(ds, Wcols, is_clf)
+                keep = np.delete(np.arange(ds.f),
+                                 np.asarray(removals[si], dtype=np.int64))
+                if is_clf:
+                    est.coef_ = est.coef_[:, keep]
+                else:
+                    est.coef_ = est.coef_[keep]
+                est.n_features_in_ = len(keep)
+                est.fit_time_ = per
+            if cluster is not None and cluster.distributed:
+                est = cluster.bcast_obj(est, src=si % cluster.world_size)
+            return est
+
+        return out, refit_fn
+
     def batched_multiclass_fit(self, X, y, cluster, mode="ovr"):
         """Train every one-vs-rest class (or one-vs-one pair) binary
         problem as one batched device solve; used by

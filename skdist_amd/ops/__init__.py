@@ -164,6 +164,12 @@ def hip_sgd_solve(ds, spec, loss_id, epochs, batch_size, seed, momentum,
     cls2_p = _pad_cols(spec.col_class2, ncp, -1)
     lr_p = _pad_cols(spec.col_lr, ncp, 0.0)
     l2_p = _pad_cols(spec.col_l2, ncp, 0.0)
+    if getattr(spec, "feat_mask", None) is not None:
+        fmask = torch.ones(fa, ncp, dtype=torch.uint8, device=device)
+        fmask[:, :ncols] = spec.feat_mask
+        fmask = fmask.contiguous()
+    else:
+        fmask = torch.empty(0, dtype=torch.uint8, device=device)
 
     # one seeded shuffle (minibatch composition then stays fixed across
     # epochs — standard for convex SGD; matches the torch reference path)
@@ -176,7 +182,7 @@ def hip_sgd_solve(ds, spec, loss_id, epochs, batch_size, seed, momentum,
         lr_scale = 1.0 / (1.0 + lr_decay * epoch)
         ext.sgd_epoch(
             Xs, XsT, GT, W, V, WbfT, partial, ys, folds,
-            cls_p, cfold_p, cls2_p, lr_p, l2_p,
+            cls_p, cfold_p, cls2_p, lr_p, l2_p, fmask,
             bs, int(loss_id), float(lr_scale), float(momentum),
             int(ds.intercept_row),
         )

@@ -10,7 +10,7 @@ extern "C" hipError_t skdist_sgd_step(
     void* GT, void* W, void* V, void* WbfT, void* partial,
     const void* y, const void* fold,
     const void* col_class, const void* col_fold, const void* col_class2,
-    const void* col_lr, const void* col_l2,
+    const void* col_lr, const void* col_l2, const void* fmask,
     long long start, long long m, long long n, long long n_pad,
     long long fa_store, int fa, int ncols_pad,
     int gt_stride, int splitk, int loss_id,
@@ -53,6 +53,7 @@ void sgd_step(torch::Tensor Xs, torch::Tensor XsT, torch::Tensor GT,
               torch::Tensor col_class, torch::Tensor col_fold,
               torch::Tensor col_class2,
               torch::Tensor col_lr, torch::Tensor col_l2,
+              torch::Tensor fmask,
               int64_t start, int64_t m, int64_t loss_id, double lr_scale,
               double momentum, int64_t intercept_row) {
     for (auto* t : {&Xs, &XsT, &GT, &W, &WbfT, &partial, &y, &fold,
@@ -68,8 +69,9 @@ void sgd_step(torch::Tensor Xs, torch::Tensor XsT, torch::Tensor GT,
         W.data_ptr(), has_V ? V.data_ptr() : nullptr, WbfT.data_ptr(),
         partial.data_ptr(), y.data_ptr(), fold.data_ptr(),
         col_class.data_ptr(), col_fold.data_ptr(), col_class2.data_ptr(),
-        col_lr.data_ptr(),
-        col_l2.data_ptr(), start, m, a.n, a.n_pad, a.fa_store, (int)a.fa,
+        col_lr.data_ptr(), col_l2.data_ptr(),
+        fmask.numel() ? fmask.data_ptr() : nullptr,
+        start, m, a.n, a.n_pad, a.fa_store, (int)a.fa,
         (int)a.ncols_pad, (int)a.gt_stride, (int)a.splitk, (int)loss_id,
         (float)lr_scale, (float)momentum, (int)intercept_row, stream);
     TORCH_CHECK(err == hipSuccess, "skdist_sgd_step: ",
@@ -84,6 +86,7 @@ void sgd_epoch(torch::Tensor Xs, torch::Tensor XsT, torch::Tensor GT,
                torch::Tensor col_class, torch::Tensor col_fold,
                torch::Tensor col_class2,
                torch::Tensor col_lr, torch::Tensor col_l2,
+               torch::Tensor fmask,
                int64_t batch_size, int64_t loss_id, double lr_scale,
                double momentum, int64_t intercept_row) {
     auto a = check_args(Xs, XsT, GT, W, partial, y);
@@ -97,8 +100,9 @@ void sgd_epoch(torch::Tensor Xs, torch::Tensor XsT, torch::Tensor GT,
             W.data_ptr(), has_V ? V.data_ptr() : nullptr, WbfT.data_ptr(),
             partial.data_ptr(), y.data_ptr(), fold.data_ptr(),
             col_class.data_ptr(), col_fold.data_ptr(),
-            col_class2.data_ptr(), col_lr.data_ptr(),
-            col_l2.data_ptr(), start, m, a.n, a.n_pad, a.fa_store,
+            col_class2.data_ptr(), col_lr.data_ptr(), col_l2.data_ptr(),
+            fmask.numel() ? fmask.data_ptr() : nullptr,
+            start, m, a.n, a.n_pad, a.fa_store,
             (int)a.fa, (int)a.ncols_pad, (int)a.gt_stride, (int)a.splitk,
             (int)loss_id, (float)lr_scale, (float)momentum,
             (int)intercept_row, stream);

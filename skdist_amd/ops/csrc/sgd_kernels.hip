@@ -271,9 +271,10 @@ extern "C" __global__ __launch_bounds__(128) void k_reduce_update(
     __bf16* __restrict__ WbfT,
     const float* __restrict__ col_lr,
     const float* __restrict__ col_l2,
-    float inv_m, float lr_scale, float momentum,
-    int intercept_row, int splitk, int fa, int ncols_pad)
-{
+    const unsigned char* __restrict__ fmask,  // [fa][ncols_pad] or null:
+    float inv_m, float lr_scale, float momentum,  // 0 pins W to 0 (the
+    int intercept_row, int splitk, int fa, int ncols_pad)  // per-column
+{                                             // feature-subset device path)
     const int col = blockIdx.x * 128 + threadIdx.x;
     const int row = blockIdx.y;
     const long long e = (long long)row * ncols_pad + col;
@@ -292,6 +293,10 @@ extern "C" __global__ __launch_bounds__(128) void k_reduce_update(
         step = v;
     }
     wv -= step;
+    if (fmask != nullptr && fmask[e] == 0) {
+        wv = 0.f;
+        if (momentum > 0.f) V[e] = 0.f;
+    }
     W[e] = wv;
     WbfT[(long long)col * fa + row] = f32_to_bf16(wv);
 }
@@ -304,7 +309,7 @@ extern "C" hipError_t skdist_sgd_step(
     void* GT, void* W, void* V, void* WbfT, void* partial,
     const void* y, const void* fold,
     const void* col_class, const void* col_fold, const void* col_class2,
-    const void* col_lr, const void* col_l2,
+    const void* col_lr, const void* col_l2, const void* fmask,
     long long start, long long m, long long n, long long n_pad,
     long long fa_store, int fa, int ncols_pad,
     int gt_stride, int splitk, int loss_id,
@@ -339,7 +344,9 @@ extern "C" hipError_t skdist_sgd_step(
         hipLaunchKernelGGL(k_reduce_update, grid, dim3(128), 0, stream,
                            (const float*)partial, (float*)W, (float*)V,
                            (__bf16*)WbfT, (const float*)col_lr,
-                           (const float*)col_l2, (float)(1.0 / (double)m),
+                           (const float*)col_l2,
+                           (const unsigned char*)fmask,
+                           (float)(1.0 / (double)m),
                            lr_scale, momentum, intercept_row, splitk, fa,
                            ncols_pad);
         HIP_CHECK(hipGetLastError());
