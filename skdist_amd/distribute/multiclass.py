@@ -227,6 +227,67 @@ class DistOneVsRestClassifier(ClassifierMixin, BaseEstimator):
     def decision_function(self, X):
         return self._scores(X)
 
+    # ------------------------------------------------------------------ #
+    # batched device inference (DistPredictor hook): all k binary models
+    # score as ONE GPU GEMM + fused sigmoid/normalize — the device
+    # replacement for the reference's O(n·k) Python assembly loop
+    # (multiclass.py:350-362; SURVEY.md §2.4 "OvR proba assembly" row)
+    # ------------------------------------------------------------------ #
+    def _stacked_linear(self):
+        coefs, inters, prob = [], [], True
+        for est in self.estimators_:
+            c = getattr(est, "coef_", None)
+            b = getattr(est, "intercept_", None)
+            if c is None or b is None:
+                return None
+            c = np.asarray(c, dtype=np.float32)
+            coefs.append(c[-1] if c.ndim == 2 else c)
+            inters.append(float(np.ravel(b)[-1]))
+            prob = prob and hasattr(est, "predict_proba")
+        return np.stack(coefs), np.asarray(inters, dtype=np.float32), prob
+
+    def _device_predict_fn(self, method, device):
+        if method not in ("predict", "predict_proba"):
+            return None
+        stacked = self._stacked_linear()
+        if stacked is None:
+            return None
+        W, b, has_proba = stacked
+        if method == "predict_proba" and not has_proba:
+            return None
+        import torch
+
+        dev = torch.device(device)
+        Wt = torch.as_tensor(W.T.copy(), device=dev)   # [f, k]
+        bt = torch.as_tensor(b, device=dev)
+
+        def fn(X, chunk=1 << 21):
+            X = np.ascontiguousarray(X, dtype=np.float32)
+            outs = []
+            for lo in range(0, len(X), chunk):
+                xb = torch.as_tensor(X[lo: lo + chunk], device=dev)
+                Z = xb @ Wt + bt
+                if has_proba:
+                    Z = torch.sigmoid(Z)
+                if method == "predict":
+                    outs.append(Z.argmax(dim=1).cpu().numpy())
+                    continue
+                if self.norm == "l1":
+                    Z = Z / Z.sum(dim=1, keepdim=True).clamp_min(1e-30)
+                elif self.norm == "l2":
+                    Z = Z / Z.norm(dim=1, keepdim=True).clamp_min(1e-30)
+                outs.append(Z.cpu().numpy())
+            out = np.concatenate(outs, axis=0)
+            if method == "predict":
+                return self.classes_[out]
+            return out
+
+        if method == "predict" and (
+            self.mlb_ is not None or getattr(self, "multilabel_", False)
+        ):
+            return None
+        return fn
+
 
 class DistOneVsOneClassifier(ClassifierMixin, BaseEstimator):
     """One-vs-one with distributed per-pair fits

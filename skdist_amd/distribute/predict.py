@@ -120,7 +120,13 @@ class DistPredictor:
                 if getattr(self.sc, "device", None) is not None
                 else "cuda"
             )
-            self._flat = flat_forest_for(self.model, device) or False
+            hook = getattr(self.model, "_device_predict_fn", None)
+            if hook is not None:
+                self._flat = hook(self.method, device) or False
+            else:
+                self._flat = flat_forest_for(self.model, device) or False
         if self._flat is False:
             return None
+        if callable(self._flat) and not hasattr(self._flat, self.method):
+            return self._flat
         return getattr(self._flat, self.method)
