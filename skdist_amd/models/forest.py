@@ -804,10 +804,138 @@ class FlatForest:
         return np.concatenate(outs, axis=0)
 
 
+def _sklearn_tree_to_hist_tree(tree, kind):
+    """Flatten one fitted sklearn tree (its ``tree_`` arrays) into the
+    HistTree layout the device traversal kernel walks.
+
+    kind: 'proba' (classifier: leaf = class distribution),
+          'value' (regressor / boosting stage: leaf = raw value).
+    """
+    t = tree.tree_
+    n = t.node_count
+    feature = t.feature.astype(np.int32).copy()
+    threshold = t.threshold.astype(np.float32)
+    left = t.children_left.astype(np.int32).copy()
+    right = t.children_right.astype(np.int32).copy()
+    leaves = np.flatnonzero(feature < 0)
+    feature[leaves] = -1
+    v = t.value[leaves]                      # [n_leaves, n_out, k]
+    if kind == "proba":
+        v = v[:, 0, :]
+        v = v / np.clip(v.sum(axis=1, keepdims=True), 1e-30, None)
+    else:
+        v = v.reshape(len(leaves), -1)[:, :1]
+    left[leaves] = np.arange(len(leaves), dtype=np.int32)
+    return HistTree(
+        feature, threshold, left, right,
+        np.ascontiguousarray(v, dtype=np.float32),
+        getattr(tree, "classes_", None),
+        t.n_features, None,
+    )
+
+
+class FlatGBT:
+    """sklearn GradientBoosting{Classifier,Regressor} flattened for the
+    device traversal kernel: raw score = prior + lr·Σ stage-tree values,
+    then the sigmoid/softmax link — the GBT batch-inference path of
+    BASELINE config 5 (reference analog: executor-side model.predict in
+    the pandas UDF, predict.py:160-178)."""
+
+    def __init__(self, model, device):
+        from sklearn.ensemble import (
+            GradientBoostingClassifier,
+            GradientBoostingRegressor,
+        )
+
+        self.is_cls = isinstance(model, GradientBoostingClassifier)
+        if not self.is_cls and not isinstance(
+                model, GradientBoostingRegressor):
+            raise TypeError(model)
+        self.classes_ = getattr(model, "classes_", None)
+        ests = model.estimators_            # [n_stages, k_trees]
+        self.k_trees = ests.shape[1]
+        lr = model.learning_rate
+        flats = []
+        for j in range(self.k_trees):
+            trees = []
+            for i in range(ests.shape[0]):
+                ht = _sklearn_tree_to_hist_tree(ests[i, j], "value")
+                ht.value = ht.value * lr
+                trees.append(ht)
+            flats.append(FlatForest(trees, device))
+        self.flats = flats
+        # prior (raw init) from the model's initial estimator via its
+        # public decision path on a single dummy row is model-dependent;
+        # use the training prior stored on the loss/init estimator
+        init = model.init_
+        probe = np.zeros((1, model.n_features_in_), dtype=np.float64)
+        if hasattr(init, "predict_proba"):
+            p = np.clip(init.predict_proba(probe)[0], 1e-12, 1 - 1e-12)
+            if len(p) == 2:
+                self.base = np.array([np.log(p[1] / p[0])])
+            else:
+                self.base = np.log(p)
+        elif hasattr(init, "predict"):
+            self.base = np.atleast_1d(
+                np.asarray(init.predict(probe), dtype=np.float64).ravel())
+        else:  # 'zero'
+            self.base = np.zeros(max(self.k_trees, 1))
+
+    def _raw(self, X):
+        cols = [
+            f.predict_value(X)[:, 0] * f.n_trees for f in self.flats
+        ]  # undo the kernel's mean -> sum
+        return np.column_stack(cols) + self.base[None, :]
+
+    def predict_proba(self, X):
+        raw = self._raw(X)
+        if raw.shape[1] == 1:
+            p1 = 1.0 / (1.0 + np.exp(-raw[:, 0]))
+            return np.column_stack([1 - p1, p1])
+        e = np.exp(raw - raw.max(axis=1, keepdims=True))
+        return e / e.sum(axis=1, keepdims=True)
+
+    def predict(self, X):
+        if self.is_cls:
+            raw = self._raw(X)
+            if raw.shape[1] == 1:
+                return self.classes_[(raw[:, 0] > 0).astype(np.int64)]
+            return self.classes_[raw.argmax(axis=1)]
+        return self._raw(X)[:, 0]
+
+
 def flat_forest_for(model, device):
-    """FlatForest for a fitted forest whose estimators_ are HistTrees,
-    else None (CPU-fitted sklearn trees keep the host path)."""
+    """Device scorer for a fitted tree ensemble: HIP-fitted HistTree
+    forests, host-fitted sklearn forests (DecisionTree estimators_), and
+    sklearn GradientBoosting models all flatten onto the same traversal
+    kernel; returns None for anything else (host path)."""
+    try:
+        from sklearn.ensemble import (
+            GradientBoostingClassifier,
+            GradientBoostingRegressor,
+        )
+
+        if isinstance(model, (GradientBoostingClassifier,
+                              GradientBoostingRegressor)):
+            return FlatGBT(model, device)
+    except Exception:
+        pass
     trees = getattr(model, "estimators_", None)
-    if not trees or not all(isinstance(t, HistTree) for t in trees):
+    if not trees:
         return None
-    return FlatForest(trees, device)
+    if all(isinstance(t, HistTree) for t in trees):
+        return FlatForest(trees, device)
+    if all(hasattr(t, "tree_") for t in trees):
+        is_cls = hasattr(trees[0], "predict_proba") and hasattr(
+            model, "classes_")
+        flat = FlatForest(
+            [
+                _sklearn_tree_to_hist_tree(
+                    t, "proba" if is_cls else "value")
+                for t in trees
+            ],
+            device,
+        )
+        flat.classes_ = getattr(model, "classes_", None)
+        return flat
+    return None

@@ -151,3 +151,23 @@ def test_bootstrap_weights_deterministic():
     assert (w1[0] != w1[1]).any()
     # multinomial: total draws == n
     assert int(w1[0].to(int).sum()) == 400
+
+
+def test_sklearn_tree_flattening_host():
+    """sklearn fitted trees flatten onto the HistTree layout with exact
+    prediction parity (host traversal; the GPU kernel walks the same
+    arrays — tested in test_forest_gpu)."""
+    from sklearn.tree import DecisionTreeClassifier, DecisionTreeRegressor
+
+    from skdist_amd.models.forest import _sklearn_tree_to_hist_tree
+
+    X, y = _cls_data(800, 6)
+    dt = DecisionTreeClassifier(max_depth=6, random_state=0).fit(X, y)
+    ht = _sklearn_tree_to_hist_tree(dt, "proba")
+    np.testing.assert_allclose(
+        ht.predict_proba(X), dt.predict_proba(X), atol=1e-6)
+
+    yr = (X[:, 0] * 2 + X[:, 1]).astype(np.float32)
+    dr = DecisionTreeRegressor(max_depth=6, random_state=0).fit(X, yr)
+    hr = _sklearn_tree_to_hist_tree(dr, "value")
+    np.testing.assert_allclose(hr.predict(X), dr.predict(X), atol=1e-5)

@@ -151,3 +151,31 @@ def test_tree_embedding_gpu():
     assert all(isinstance(t, HistTree) for t in emb.estimators_)
     T2 = emb.transform(X)
     assert (T != T2).nnz == 0
+
+
+def test_sklearn_forest_and_gbt_on_device():
+    """Host-fitted sklearn ensembles score through the traversal kernel
+    (BASELINE config 5's GBT batch-inference path)."""
+    from sklearn.ensemble import (
+        GradientBoostingClassifier,
+        RandomForestClassifier,
+    )
+
+    from skdist_amd.models.forest import flat_forest_for
+
+    X, y = _cls_data(3000, 10, seed=9)
+    rf = RandomForestClassifier(n_estimators=20, random_state=0).fit(X, y)
+    ff = flat_forest_for(rf, "cuda")
+    np.testing.assert_allclose(
+        ff.predict_proba(X), rf.predict_proba(X), atol=1e-5)
+
+    gbt = GradientBoostingClassifier(
+        n_estimators=25, random_state=0).fit(X, y)
+    fg = flat_forest_for(gbt, "cuda")
+    np.testing.assert_allclose(
+        fg.predict_proba(X), gbt.predict_proba(X), atol=1e-4)
+    assert (fg.predict(X) == gbt.predict(X)).mean() > 0.999
+
+    pred = DistPredictor(gbt, sc=None, method="predict_proba")
+    np.testing.assert_allclose(
+        pred(X), gbt.predict_proba(X), atol=1e-4)
