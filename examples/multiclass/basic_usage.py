@@ -1,0 +1,34 @@
+"""
+DistOneVsRestClassifier / DistOneVsOneClassifier (reference analog:
+examples/multiclass/basic_usage.py on digits; the reference reports OvR
+weighted F1 0.9589).
+
+Every per-class (or per-pair) binary problem trains as one column of a
+single batched solve; ``predict_proba`` normalizes the k binary columns
+(norm='l1'|'l2'), replacing the reference's O(n*k) Python loop
+(multiclass.py:350-362) with one vectorized pass (GPU GEMM via
+DistPredictor).
+"""
+
+import numpy as np
+from sklearn.datasets import load_digits
+from sklearn.metrics import f1_score
+
+from skdist_amd.distribute.multiclass import (
+    DistOneVsOneClassifier,
+    DistOneVsRestClassifier,
+)
+from skdist_amd.models import LogisticRegression
+
+X, y = load_digits(return_X_y=True)
+X = np.asarray(X, dtype=np.float32)
+
+ovr = DistOneVsRestClassifier(
+    LogisticRegression(epochs=30, random_state=0), norm="l1", sc=None)
+ovr.fit(X, y)
+print("OvR F1w:", round(f1_score(y, ovr.predict(X), average="weighted"), 4))
+
+ovo = DistOneVsOneClassifier(
+    LogisticRegression(epochs=30, random_state=0), sc=None)
+ovo.fit(X, y)
+print("OvO F1w:", round(f1_score(y, ovo.predict(X), average="weighted"), 4))
