@@ -127,10 +127,11 @@ class DeviceDataset:
             mean = Xt.mean(dim=0)
             std = Xt.std(dim=0, unbiased=False)
             std = torch.where(std > 1e-12, std, torch.ones_like(std))
-            Xt = (Xt - mean) / std
             self.feat_mean = mean.cpu().numpy()
             self.feat_std = std.cpu().numpy()
         else:
+            mean = torch.zeros(self.f, device=self.device)
+            std = torch.ones(self.f, device=self.device)
             self.feat_mean = np.zeros(self.f, dtype=np.float32)
             self.feat_std = np.ones(self.f, dtype=np.float32)
         # [X | 1 | 0-pad]: the ones column is the intercept; on GPU the
@@ -140,12 +141,23 @@ class DeviceDataset:
         fa = self.f + 1
         pad = (-fa) % 32 if self.device.type == "cuda" else 0
         self.fa = fa + pad
-        cols = [Xt, torch.ones(self.n, 1, dtype=Xt.dtype, device=self.device)]
-        if pad:
-            cols.append(
-                torch.zeros(self.n, pad, dtype=Xt.dtype, device=self.device)
-            )
-        self.Xaug = torch.cat(cols, dim=1).to(comp_dtype).contiguous()
+        if _use_hip(self.device) and self.fa % 8 == 0:
+            # fused standardize + augment + bf16 (one pass, K4 kernel)
+            from ..ops import require_hip
+
+            self.Xaug = torch.empty(self.n, self.fa, dtype=comp_dtype,
+                                    device=self.device)
+            require_hip().standardize(
+                Xt.contiguous(), mean.contiguous(),
+                (1.0 / std).contiguous(), self.Xaug, self.f)
+        else:
+            Xt = (Xt - mean) / std
+            cols = [Xt, torch.ones(self.n, 1, dtype=Xt.dtype,
+                                   device=self.device)]
+            if pad:
+                cols.append(torch.zeros(self.n, pad, dtype=Xt.dtype,
+                                        device=self.device))
+            self.Xaug = torch.cat(cols, dim=1).to(comp_dtype).contiguous()
         del Xt
 
         self.fold_id = None  # set by set_cv_partition

@@ -144,6 +144,9 @@ extern "C" hipError_t skdist_forest_apply(
     const void* X, const void* feat, const void* thr, const void* left,
     const void* right, const void* roots, void* out_leaf, long long rows,
     int f, int n_trees, hipStream_t stream);
+extern "C" hipError_t skdist_standardize(
+    const void* X, const void* mean, const void* inv_std, void* out,
+    long long n, int f, int fa, hipStream_t stream);
 extern "C" hipError_t skdist_hash_vectorize(
     const void* bytes, const void* doc_off, long long n_docs, int mode,
     int min_n, int max_n, int n_features, int alt_sign, void* out_keys,
@@ -250,6 +253,21 @@ void forest_apply(torch::Tensor X, torch::Tensor feat, torch::Tensor thr,
                 hipGetErrorString(err));
 }
 
+void standardize(torch::Tensor X, torch::Tensor mean,
+                 torch::Tensor inv_std, torch::Tensor out, int64_t f) {
+    for (auto* t : {&X, &mean, &inv_std, &out}) {
+        CHECK_DEV(*t);
+        CHECK_CONT(*t);
+    }
+    TORCH_CHECK(out.scalar_type() == torch::kBFloat16, "out must be bf16");
+    auto stream = c10::hip::getCurrentHIPStream().stream();
+    hipError_t err = skdist_standardize(
+        X.data_ptr(), mean.data_ptr(), inv_std.data_ptr(), out.data_ptr(),
+        out.size(0), (int)f, (int)out.size(1), stream);
+    TORCH_CHECK(err == hipSuccess, "standardize: ",
+                hipGetErrorString(err));
+}
+
 void hash_vectorize(torch::Tensor bytes, torch::Tensor doc_off,
                     int64_t mode, int64_t min_n, int64_t max_n,
                     int64_t n_features, int64_t alt_sign,
@@ -275,6 +293,8 @@ void hash_vectorize(torch::Tensor bytes, torch::Tensor doc_off,
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("sgd_step", &sgd_step, "fused batched SGD step (K1+K2+K3)");
     m.def("sgd_epoch", &sgd_epoch, "one epoch of fused SGD steps");
+    m.def("standardize", &standardize,
+          "fused standardize + augment + bf16 cast");
     m.def("hash_vectorize", &hash_vectorize,
           "tokenize + murmur3 feature hashing -> COO pairs");
     m.def("tree_hist", &tree_hist, "per-(node,feature,bin) stats");

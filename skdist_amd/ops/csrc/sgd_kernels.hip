@@ -353,3 +353,54 @@ extern "C" hipError_t skdist_sgd_step(
     }
     return hipSuccess;
 }
+
+// ---------------------------------------------------------------------- //
+// K4: fused standardize + intercept/pad augment + bf16 cast
+// out[i][j] = (X[i][j] - mean[j]) * inv_std[j]   for j <  f
+//           = 1.0                                 for j == f (intercept)
+//           = 0.0                                 for j >  f (K-pad)
+// One pass over X instead of the eager sub/div/cat/cast chain
+// (DeviceDataset build, skdist_amd/models/_sgd.py).
+// grid: ceil(n*fa/8/256), block 256; each thread emits one 16 B chunk.
+// ---------------------------------------------------------------------- //
+extern "C" __global__ __launch_bounds__(256) void k_standardize(
+    const float* __restrict__ X, const float* __restrict__ mean,
+    const float* __restrict__ inv_std, __bf16* __restrict__ out,
+    long long n, int f, int fa)
+{
+    const long long chunk =
+        (long long)blockIdx.x * blockDim.x + threadIdx.x;
+    const long long total = n * (long long)fa / 8;
+    if (chunk >= total) return;
+    const long long e0 = chunk * 8;
+    const long long row = e0 / fa;
+    const int c0 = (int)(e0 - row * fa);
+    const float* xr = X + row * (long long)f;
+    short4v lo, hi;
+    #pragma unroll
+    for (int j = 0; j < 8; ++j) {
+        const int c = c0 + j;
+        float v;
+        if (c < f)
+            v = (xr[c] - mean[c]) * inv_std[c];
+        else
+            v = (c == f) ? 1.f : 0.f;
+        const short b = __builtin_bit_cast(short, f32_to_bf16(v));
+        if (j < 4) lo[j] = b; else hi[j - 4] = b;
+    }
+    *(short4v*)(out + e0) = lo;
+    *(short4v*)(out + e0 + 4) = hi;
+}
+
+extern "C" hipError_t skdist_standardize(
+    const void* X, const void* mean, const void* inv_std, void* out,
+    long long n, int f, int fa, hipStream_t stream)
+{
+    if (fa % 8 != 0) return hipErrorInvalidValue;
+    const long long chunks = n * (long long)fa / 8;
+    const int blocks = (int)((chunks + 255) / 256);
+    hipLaunchKernelGGL(k_standardize, dim3(blocks), dim3(256), 0, stream,
+                       (const float*)X, (const float*)mean,
+                       (const float*)inv_std, (__bf16*)out, n, f, fa);
+    return hipGetLastError();
+}
