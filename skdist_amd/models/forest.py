@@ -6,7 +6,8 @@ Replaces the per-tree sklearn Cython builder the reference fans out
 inventory SURVEY.md §2.4 row 2) with a level-synchronous GPU builder:
 
   * X is quantile-binned ONCE into uint8 codes resident in HBM
-    (feature-major ``[f, n]`` so gathered row reads coalesce);
+    (row-major ``[n, fp]``, feature stride padded to a multiple of 4 so
+    the histogram kernel reads 4 codes per uchar4 load);
   * a batch of TB trees grows together, breadth-first: one fused
     LDS-staged histogram kernel per level covers every (tree, node)
     frontier entry, a split kernel scans all (feature, bin) candidates
@@ -164,11 +165,17 @@ class BinnedDataset:
         edges = torch.quantile(Xs.to(torch.float32), qs, dim=0)  # [nb-1, f]
         self.edges = edges.t().contiguous()                      # [f, nb-1]
 
-        # codes[f][n]: count of edges < x  →  (code <= b) ⟺ (x <= edges[b])
+        # codes[n][fp]: count of edges < x  →  (code <= b) ⟺ (x <= edges[b])
+        # row-major with the feature stride padded to a multiple of 4 so
+        # the histogram kernel reads 4 feature codes per uchar4 load
+        self.fp = (self.f + 3) // 4 * 4
         XT = Xt.t().contiguous()
-        codes = torch.searchsorted(self.edges, XT, right=False)
-        self.codes = codes.to(torch.uint8).contiguous()
-        del XT, Xt, Xs
+        codes_fm = torch.searchsorted(self.edges, XT, right=False)
+        self.codes = torch.zeros(self.n, self.fp, dtype=torch.uint8,
+                                 device=self.device)
+        self.codes[:, : self.f] = codes_fm.to(torch.uint8).t()
+        self.codes = self.codes.contiguous()
+        del XT, Xt, Xs, codes_fm
 
         self.is_cls = is_cls
         if is_cls:
@@ -243,7 +250,8 @@ class ForestBuilder:
         self.engine = engine
         # features per LDS group for the hist kernel
         per_feat = ds.nbins * ds.S * 4
-        self.fg = max(1, min(ds.f, self.LDS_BUDGET_BYTES // per_feat))
+        fg = max(1, min(ds.f, self.LDS_BUDGET_BYTES // per_feat))
+        self.fg = fg // 4 * 4 if fg >= 4 else fg
 
     # -------------------------------------------------------------- #
     def build(self, seeds, sample_weight=None):
@@ -545,7 +553,7 @@ class ForestBuilder:
             rows = si[t, int(fr_start[k]): int(fr_start[k] + fr_count[k])]
             rows = rows.to(torch.int64)
             w = weights[t, rows].to(torch.float32)
-            codes = ds.codes[:, rows].to(torch.int64)          # [f, m]
+            codes = ds.codes[rows][:, : f].t().to(torch.int64)  # [f, m]
             hist = torch.zeros(f, nbins, S, device=ds.device)
             if ds.is_cls:
                 stat = ds.y_int[rows].to(torch.int64)          # [m]
@@ -663,7 +671,7 @@ class ForestBuilder:
                 t = int(p_tree[k])
                 st, cnt = int(p_start[k]), int(p_count[k])
                 rows = si_in[t, st:st + cnt].to(torch.int64)
-                go_left = ds.codes[int(p_feat[k]), rows] <= int(p_bin[k])
+                go_left = ds.codes[rows, int(p_feat[k])] <= int(p_bin[k])
                 lrows = rows[go_left]
                 rrows = rows[~go_left]
                 nl[k] = len(lrows)

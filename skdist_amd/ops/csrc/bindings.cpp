@@ -119,8 +119,8 @@ void sgd_epoch(torch::Tensor Xs, torch::Tensor XsT, torch::Tensor GT,
 extern "C" hipError_t skdist_tree_hist(
     const void* codes, const void* y_int, const void* y_f,
     const void* weights, const void* sample_idx, const void* chunks,
-    void* hist, long long n, int f, int nbins, int S, int is_cls, int fg,
-    int n_chunks, hipStream_t stream);
+    void* hist, long long n, int f, int fp, int nbins, int S, int is_cls,
+    int fg, int n_chunks, hipStream_t stream);
 extern "C" hipError_t skdist_tree_split(
     const void* hist, const void* node_seed, int n_frontier, int f,
     int nbins, int S, int is_cls, int crit, int m_features, int extra_mode,
@@ -129,12 +129,12 @@ extern "C" hipError_t skdist_tree_split(
     hipStream_t stream);
 extern "C" hipError_t skdist_part_count(
     const void* codes, const void* sample_idx, const void* chunks,
-    const void* split_feat, const void* split_bin, long long n,
+    const void* split_feat, const void* split_bin, long long n, int fp,
     int n_chunks, void* out_counts, hipStream_t stream);
 extern "C" hipError_t skdist_part_scatter(
     const void* codes, const void* sample_idx_in, const void* chunks,
     const void* split_feat, const void* split_bin, const void* left_base,
-    const void* right_base, long long n, int n_chunks,
+    const void* right_base, long long n, int fp, int n_chunks,
     void* sample_idx_out, hipStream_t stream);
 extern "C" hipError_t skdist_forest_predict(
     const void* X, const void* feat, const void* thr, const void* left,
@@ -159,6 +159,7 @@ void tree_hist(torch::Tensor codes, torch::Tensor y_int, torch::Tensor y_f,
                torch::Tensor chunks, torch::Tensor hist, int64_t n,
                int64_t f, int64_t nbins, int64_t S, int64_t is_cls,
                int64_t fg) {
+    const int64_t fp = codes.size(1);
     for (auto* t : {&codes, &weights, &sample_idx, &chunks, &hist}) {
         CHECK_DEV(*t);
         CHECK_CONT(*t);
@@ -170,7 +171,7 @@ void tree_hist(torch::Tensor codes, torch::Tensor y_int, torch::Tensor y_f,
         codes.data_ptr(), is_cls ? y_int.data_ptr() : nullptr,
         is_cls ? nullptr : y_f.data_ptr(), weights.data_ptr(),
         sample_idx.data_ptr(), chunks.data_ptr(), hist.data_ptr(), n,
-        (int)f, (int)nbins, (int)S, (int)is_cls, (int)fg,
+        (int)f, (int)fp, (int)nbins, (int)S, (int)is_cls, (int)fg,
         (int)chunks.size(0), stream);
     TORCH_CHECK(err == hipSuccess, "tree_hist: ", hipGetErrorString(err));
 }
@@ -204,7 +205,8 @@ void part_count(torch::Tensor codes, torch::Tensor sample_idx,
     hipError_t err = skdist_part_count(
         codes.data_ptr(), sample_idx.data_ptr(), chunks.data_ptr(),
         split_feat.data_ptr(), split_bin.data_ptr(), n,
-        (int)chunks.size(0), out_counts.data_ptr(), stream);
+        (int)codes.size(1), (int)chunks.size(0), out_counts.data_ptr(),
+        stream);
     TORCH_CHECK(err == hipSuccess, "part_count: ", hipGetErrorString(err));
 }
 
@@ -217,8 +219,8 @@ void part_scatter(torch::Tensor codes, torch::Tensor sample_idx_in,
     hipError_t err = skdist_part_scatter(
         codes.data_ptr(), sample_idx_in.data_ptr(), chunks.data_ptr(),
         split_feat.data_ptr(), split_bin.data_ptr(), left_base.data_ptr(),
-        right_base.data_ptr(), n, (int)chunks.size(0),
-        sample_idx_out.data_ptr(), stream);
+        right_base.data_ptr(), n, (int)codes.size(1),
+        (int)chunks.size(0), sample_idx_out.data_ptr(), stream);
     TORCH_CHECK(err == hipSuccess, "part_scatter: ",
                 hipGetErrorString(err));
 }

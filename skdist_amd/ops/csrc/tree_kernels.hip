@@ -15,7 +15,8 @@
 // Host orchestration + torch eager reference: skdist_amd/models/forest.py.
 //
 // Layouts:
-//   codes      [f][n]    uint8   feature-major quantile bin codes
+//   codes      [n][fp]   uint8   row-major quantile bin codes (fp = f
+//                                     padded to a multiple of 4 for uchar4 loads)
 //   sample_idx [TB][n]   int32   per-tree row ids, node segments contiguous
 //   weights    [TB][n]   uint8   bootstrap multiplicities (0 = out of bag)
 //   hist       [NF][f][nbins][S] f32   S = n_classes (cls) | 3 (w,wy,wyy)
@@ -44,15 +45,17 @@ static __device__ __forceinline__ unsigned wang_hash(unsigned s) {
 // grid: (n_chunks, n_feature_groups), block 256
 // dynamic LDS: fg * nbins * S floats
 // ---------------------------------------------------------------------- //
+typedef __attribute__((ext_vector_type(4))) unsigned char uchar4v;
+
 extern "C" __global__ __launch_bounds__(256) void k_tree_hist(
-    const unsigned char* __restrict__ codes,    // [f][n]
+    const unsigned char* __restrict__ codes,    // [n][fp] row-major
     const int* __restrict__ y_int,              // [n] (cls) or nullptr
     const float* __restrict__ y_f,              // [n] (reg) or nullptr
     const unsigned char* __restrict__ weights,  // [TB][n]
     const int* __restrict__ sample_idx,         // [TB][n]
     const int* __restrict__ chunks,             // [n_chunks][4]
     float* __restrict__ hist,                   // [NF][f][nbins][S]
-    long long n, int f, int nbins, int S, int is_cls, int fg) {
+    long long n, int f, int fp, int nbins, int S, int is_cls, int fg) {
     extern __shared__ float lds[];
     const int chunk = blockIdx.x;
     const int g0 = blockIdx.y * fg;
@@ -68,6 +71,9 @@ extern "C" __global__ __launch_bounds__(256) void k_tree_hist(
 
     const int* si = sample_idx + (long long)tree_slot * n + row_start;
     const unsigned char* wrow = weights + (long long)tree_slot * n;
+    // vector path: 4 feature codes per uchar4 load (codes rows are
+    // padded to fp, a multiple of 4, so over-reads land in zero pad)
+    const bool vec = (g0 & 3) == 0;
     for (int r = threadIdx.x; r < row_count; r += blockDim.x) {
         const int i = si[r];
         const float w = (float)wrow[i];
@@ -80,15 +86,35 @@ extern "C" __global__ __launch_bounds__(256) void k_tree_hist(
             v1 = w * yv;
             v2 = w * yv * yv;
         }
-        for (int jj = 0; jj < nf_g; ++jj) {
-            const int b = codes[(long long)(g0 + jj) * n + i];
-            float* h = lds + ((jj * nbins + b) * S);
-            if (is_cls) {
-                atomicAdd(h + stat, w);
-            } else {
-                atomicAdd(h + 0, w);
-                atomicAdd(h + 1, v1);
-                atomicAdd(h + 2, v2);
+        const unsigned char* crow = codes + (long long)i * fp + g0;
+        if (vec) {
+            const uchar4v* c4 = (const uchar4v*)crow;
+            for (int q = 0; q * 4 < nf_g; ++q) {
+                const uchar4v b4 = c4[q];
+                #pragma unroll
+                for (int t = 0; t < 4; ++t) {
+                    const int jj = q * 4 + t;
+                    if (jj >= nf_g) break;
+                    float* h = lds + ((jj * nbins + (int)b4[t]) * S);
+                    if (is_cls) {
+                        atomicAdd(h + stat, w);
+                    } else {
+                        atomicAdd(h + 0, w);
+                        atomicAdd(h + 1, v1);
+                        atomicAdd(h + 2, v2);
+                    }
+                }
+            }
+        } else {
+            for (int jj = 0; jj < nf_g; ++jj) {
+                float* h = lds + ((jj * nbins + (int)crow[jj]) * S);
+                if (is_cls) {
+                    atomicAdd(h + stat, w);
+                } else {
+                    atomicAdd(h + 0, w);
+                    atomicAdd(h + 1, v1);
+                    atomicAdd(h + 2, v2);
+                }
             }
         }
     }
@@ -323,15 +349,14 @@ extern "C" __global__ __launch_bounds__(256) void k_tree_split(
 extern "C" __global__ __launch_bounds__(256) void k_part_count(
     const unsigned char* __restrict__ codes, const int* __restrict__ sample_idx,
     const int* __restrict__ chunks, const int* __restrict__ split_feat,
-    const int* __restrict__ split_bin, long long n,
+    const int* __restrict__ split_bin, long long n, int fp,
     int* __restrict__ out_counts) {
     const int chunk = blockIdx.x;
     const int part_slot = chunks[chunk * 4 + 0];
     const int tree_slot = chunks[chunk * 4 + 1];
     const int row_start = chunks[chunk * 4 + 2];
     const int row_count = chunks[chunk * 4 + 3];
-    const unsigned char* cj =
-        codes + (long long)split_feat[part_slot] * n;
+    const int jf = split_feat[part_slot];
     const int b = split_bin[part_slot];
     const int* si = sample_idx + (long long)tree_slot * n + row_start;
     __shared__ int s_cnt;
@@ -339,7 +364,7 @@ extern "C" __global__ __launch_bounds__(256) void k_part_count(
     __syncthreads();
     int cnt = 0;
     for (int r = threadIdx.x; r < row_count; r += blockDim.x)
-        if (cj[si[r]] <= b) ++cnt;
+        if (codes[(long long)si[r] * fp + jf] <= b) ++cnt;
     atomicAdd(&s_cnt, cnt);
     __syncthreads();
     if (threadIdx.x == 0) out_counts[chunk] = s_cnt;
@@ -355,14 +380,13 @@ extern "C" __global__ __launch_bounds__(256) void k_part_scatter(
     const int* __restrict__ sample_idx_in, const int* __restrict__ chunks,
     const int* __restrict__ split_feat, const int* __restrict__ split_bin,
     const int* __restrict__ left_base, const int* __restrict__ right_base,
-    long long n, int* __restrict__ sample_idx_out) {
+    long long n, int fp, int* __restrict__ sample_idx_out) {
     const int chunk = blockIdx.x;
     const int part_slot = chunks[chunk * 4 + 0];
     const int tree_slot = chunks[chunk * 4 + 1];
     const int row_start = chunks[chunk * 4 + 2];
     const int row_count = chunks[chunk * 4 + 3];
-    const unsigned char* cj =
-        codes + (long long)split_feat[part_slot] * n;
+    const int jf = split_feat[part_slot];
     const int b = split_bin[part_slot];
     const int* si = sample_idx_in + (long long)tree_slot * n + row_start;
     int* so = sample_idx_out + (long long)tree_slot * n;
@@ -382,7 +406,7 @@ extern "C" __global__ __launch_bounds__(256) void k_part_scatter(
         bool p = false;
         if (valid) {
             i = si[r];
-            p = cj[i] <= b;
+            p = codes[(long long)i * fp + jf] <= b;
         }
         const unsigned long long mask = __ballot(p);
         const int before =
@@ -417,8 +441,8 @@ extern "C" __global__ __launch_bounds__(256) void k_part_scatter(
 extern "C" hipError_t skdist_tree_hist(
     const void* codes, const void* y_int, const void* y_f,
     const void* weights, const void* sample_idx, const void* chunks,
-    void* hist, long long n, int f, int nbins, int S, int is_cls, int fg,
-    int n_chunks, hipStream_t stream) {
+    void* hist, long long n, int f, int fp, int nbins, int S, int is_cls,
+    int fg, int n_chunks, hipStream_t stream) {
     const int n_groups = (f + fg - 1) / fg;
     const size_t lds = (size_t)fg * nbins * S * sizeof(float);
     hipLaunchKernelGGL(k_tree_hist, dim3(n_chunks, n_groups), dim3(256),
@@ -426,7 +450,7 @@ extern "C" hipError_t skdist_tree_hist(
                        (const int*)y_int, (const float*)y_f,
                        (const unsigned char*)weights,
                        (const int*)sample_idx, (const int*)chunks,
-                       (float*)hist, n, f, nbins, S, is_cls, fg);
+                       (float*)hist, n, f, fp, nbins, S, is_cls, fg);
     return hipGetLastError();
 }
 
@@ -448,25 +472,25 @@ extern "C" hipError_t skdist_tree_split(
 
 extern "C" hipError_t skdist_part_count(
     const void* codes, const void* sample_idx, const void* chunks,
-    const void* split_feat, const void* split_bin, long long n,
+    const void* split_feat, const void* split_bin, long long n, int fp,
     int n_chunks, void* out_counts, hipStream_t stream) {
     hipLaunchKernelGGL(k_part_count, dim3(n_chunks), dim3(256), 0, stream,
                        (const unsigned char*)codes, (const int*)sample_idx,
                        (const int*)chunks, (const int*)split_feat,
-                       (const int*)split_bin, n, (int*)out_counts);
+                       (const int*)split_bin, n, fp, (int*)out_counts);
     return hipGetLastError();
 }
 
 extern "C" hipError_t skdist_part_scatter(
     const void* codes, const void* sample_idx_in, const void* chunks,
     const void* split_feat, const void* split_bin, const void* left_base,
-    const void* right_base, long long n, int n_chunks,
+    const void* right_base, long long n, int fp, int n_chunks,
     void* sample_idx_out, hipStream_t stream) {
     hipLaunchKernelGGL(k_part_scatter, dim3(n_chunks), dim3(256), 0,
                        stream, (const unsigned char*)codes,
                        (const int*)sample_idx_in, (const int*)chunks,
                        (const int*)split_feat, (const int*)split_bin,
                        (const int*)left_base, (const int*)right_base, n,
-                       (int*)sample_idx_out);
+                       fp, (int*)sample_idx_out);
     return hipGetLastError();
 }

@@ -47,7 +47,7 @@ def test_binning_roundtrip():
     codes = ds.codes.numpy()
     for j in range(4):
         for b in (0, 7, 14):
-            lhs = codes[j] <= b
+            lhs = codes[:, j] <= b
             rhs = X[:, j] <= edges[j, b]
             assert (lhs == rhs).all()
 
