@@ -1,21 +1,26 @@
 export TMPDIR=/tmp
 mkdir -p /root/repo/gpurun_out
 cd /root/repo
-echo "== numerics with BN=256 kernels =="
-timeout 600 python -m pytest tests/test_gpu_kernels.py -x -q > gpurun_out/pytest_k.log 2>&1; echo "rc=$?"
-tail -2 gpurun_out/pytest_k.log
-run_bench () {
-  cp skdist_amd/ops/variants/$1 skdist_amd/ops/_skdist_hip.so
-  timeout 700 python bench.py --steps 3 --warmup 1 2>/dev/null | tail -1 | python3 -c "import json,sys; d=json.load(sys.stdin); print('$1', round(d['ms_per_step'],1), 'ms/step', round(d['value']), 'fits/s')"
-}
-echo "== same-box A/B 128 vs 256 tiles =="
-run_bench hip_bn128.so
-run_bench hip_bn256.so
-run_bench hip_bn128.so
-run_bench hip_bn256.so
-cp skdist_amd/ops/variants/hip_bn256.so skdist_amd/ops/_skdist_hip.so
-echo "== stats for BN=256 =="
-cd /tmp
-timeout 600 rocprofv3 --kernel-trace --stats -d /root/repo/gpurun_out/prof5 -o prof5 -- python /root/repo/bench.py --steps 1 --warmup 1 --epochs 6 > /root/repo/gpurun_out/prof5.log 2>&1
-cd /root/repo
-python tools/prof_summary.py gpurun_out/prof5/prof5_results.db 2>&1 | head -6
+echo "== forest gpu tests =="
+timeout 600 python -m pytest tests/test_forest_gpu.py -x -q 2>&1 | tail -2
+echo "== forest perf (same probe as r01: 32/64 trees depth12, 1M x 64) =="
+PYTHONPATH=/root/repo timeout 600 python - <<'PY' 2>&1 | tail -4
+import numpy as np, time, torch
+from skdist_amd.models.forest import BinnedDataset, ForestBuilder, FlatForest
+rng = np.random.default_rng(0)
+n, f = 1_000_000, 64
+X = rng.standard_normal((n, f)).astype(np.float32)
+w = rng.standard_normal(f)
+y = ((X @ w + 0.3*rng.standard_normal(n)) > 0).astype(np.int64)
+ds = BinnedDataset(X, y, "cuda", is_cls=True)
+b = ForestBuilder(ds, "gini", max_depth=12, max_features="sqrt", bootstrap=True, tree_batch=32)
+b.build([0])  # warm
+torch.cuda.synchronize(); t0 = time.time()
+trees = b.build(list(range(64)))
+torch.cuda.synchronize(); dt = time.time()-t0
+print(f"64 trees depth12: {dt:.2f}s = {64/dt:.2f} trees/s (was 26.07)")
+ff = FlatForest(trees, "cuda")
+t0=time.time(); p = ff.predict_value(X); torch.cuda.synchronize()
+print(f"predict 1M x 64 trees: {time.time()-t0:.3f}s")
+print("acc", (ds.classes_[p.argmax(1)] == y).mean())
+PY
