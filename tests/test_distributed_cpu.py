@@ -26,6 +26,9 @@ def _spmd_worker(rank, world_size, port, path, q):
     os.environ["MASTER_PORT"] = str(port)
     os.environ["RANK"] = str(rank)
     os.environ["WORLD_SIZE"] = str(world_size)
+    # CPU/gloo plumbing tests even on a GPU box (two ranks cannot share
+    # the single visible device over NCCL)
+    os.environ["CUDA_VISIBLE_DEVICES"] = ""
     try:
         result = _SCENARIOS[path](rank)
         q.put((rank, "ok", result))
@@ -87,9 +90,104 @@ def _scenario_generic(rank):
     }
 
 
+def _scenario_forest(rank):
+    from skdist_amd import Cluster
+    from skdist_amd.distribute.ensemble import DistRandomForestClassifier
+
+    sc = Cluster()
+    X = y = None
+    if rank == 0:
+        rng = np.random.default_rng(0)
+        X = rng.standard_normal((600, 8)).astype(np.float32)
+        y = (X[:, 0] + X[:, 1] > 0).astype(np.int64)
+    clf = DistRandomForestClassifier(
+        sc=sc, n_estimators=12, random_state=0)
+    clf.fit(X, y)
+    Xh = sc.sync_host_data(X)  # for the local score below
+    proba = clf.predict_proba(Xh)
+    assert len(pickle.dumps(clf)) > 0  # sc-free, picklable
+    return {
+        "n_trees": len(clf.estimators_),
+        "acc": float((clf.predict(Xh) == sc.sync_host_data(y)).mean()),
+        "proba_sum": float(proba[:, 1].sum()),
+    }
+
+
+def _scenario_ovr(rank):
+    from skdist_amd import Cluster
+    from skdist_amd.distribute.multiclass import DistOneVsRestClassifier
+    from skdist_amd.models import LogisticRegression
+
+    sc = Cluster()
+    X = y = None
+    if rank == 0:
+        rng = np.random.default_rng(1)
+        X = rng.standard_normal((900, 10)).astype(np.float32)
+        W = rng.standard_normal((4, 10))
+        y = (X @ W.T).argmax(axis=1)
+    ovr = DistOneVsRestClassifier(
+        LogisticRegression(epochs=10, random_state=0), norm="l1", sc=sc)
+    ovr.fit(X, y)
+    Xh = sc.sync_host_data(X)
+    proba = ovr.predict_proba(Xh)
+    return {
+        "k": len(ovr.estimators_),
+        "proba0": [float(v) for v in proba[0]],
+        "acc": float((ovr.predict(Xh) == sc.sync_host_data(y)).mean()),
+    }
+
+
+def _scenario_eliminate(rank):
+    from skdist_amd import Cluster
+    from skdist_amd.distribute.eliminate import DistFeatureEliminator
+    from skdist_amd.models import LogisticRegression
+
+    sc = Cluster()
+    X = y = None
+    if rank == 0:
+        rng = np.random.default_rng(2)
+        X = rng.standard_normal((1500, 10)).astype(np.float32)
+        w = np.zeros(10)
+        w[:5] = rng.standard_normal(5) * 2
+        y = ((X @ w + 0.2 * rng.standard_normal(1500)) > 0).astype(
+            np.int64)
+        X[:, 5:] = rng.standard_normal((1500, 5))
+    el = DistFeatureEliminator(
+        LogisticRegression(epochs=10, random_state=0), sc=sc,
+        min_features_to_select=3, step=2, cv=3)
+    el.fit(X, y)
+    return {
+        "kept": [int(v) for v in el.best_features_],
+        "scores": [float(s) for s in el.scores_],
+    }
+
+
+def _scenario_predictor(rank):
+    from sklearn.ensemble import RandomForestClassifier
+
+    from skdist_amd import Cluster
+    from skdist_amd.distribute.predict import DistPredictor
+
+    sc = Cluster()
+    rng = np.random.default_rng(3)
+    X = rng.standard_normal((2000, 6)).astype(np.float32)
+    y = (X[:, 0] > 0).astype(int)
+    model = RandomForestClassifier(n_estimators=10, random_state=0)
+    model.fit(X[:500], y[:500])
+    pred = DistPredictor(model, sc=sc, method="predict_proba",
+                         chunk_rows=300)
+    out = pred(X)  # sharded across ranks, gathered everywhere
+    ref = model.predict_proba(X)
+    return {"match": bool(np.allclose(out, ref)), "n": len(out)}
+
+
 _SCENARIOS = {
     "batched": _scenario_batched,
     "generic": _scenario_generic,
+    "forest": _scenario_forest,
+    "ovr": _scenario_ovr,
+    "eliminate": _scenario_eliminate,
+    "predictor": _scenario_predictor,
 }
 
 
@@ -128,3 +226,35 @@ def test_spmd_generic_gloo():
     outs = _run_spmd("generic")
     assert outs[0]["best_score"] > 0.9
     assert np.allclose(outs[0]["scores"], outs[1]["scores"])
+
+
+@pytest.mark.timeout(300)
+def test_spmd_forest_gloo():
+    outs = _run_spmd("forest")
+    assert outs[0]["n_trees"] == 12
+    assert outs[0]["acc"] > 0.9
+    # both ranks hold the identical fitted forest
+    assert np.isclose(outs[0]["proba_sum"], outs[1]["proba_sum"])
+
+
+@pytest.mark.timeout(300)
+def test_spmd_ovr_gloo():
+    outs = _run_spmd("ovr")
+    assert outs[0]["k"] == 4
+    assert outs[0]["acc"] > 0.85
+    assert np.allclose(outs[0]["proba0"], outs[1]["proba0"])
+
+
+@pytest.mark.timeout(300)
+def test_spmd_eliminate_gloo():
+    outs = _run_spmd("eliminate")
+    assert outs[0]["kept"] == outs[1]["kept"]
+    assert set(outs[0]["kept"]) == set(range(5))
+    assert np.allclose(outs[0]["scores"], outs[1]["scores"])
+
+
+@pytest.mark.timeout(300)
+def test_spmd_predictor_gloo():
+    outs = _run_spmd("predictor")
+    assert outs[0]["match"] and outs[1]["match"]
+    assert outs[0]["n"] == 2000
