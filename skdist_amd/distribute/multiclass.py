@@ -383,6 +383,54 @@ class DistOneVsOneClassifier(ClassifierMixin, BaseEstimator):
             conf[:, i] -= d
         return conf
 
+    # batched device inference (DistPredictor hook): all k(k-1)/2 pair
+    # models score as ONE GPU GEMM, votes/confidences fused on device
+    def _device_predict_fn(self, method, device):
+        if method != "predict":
+            return None
+        coefs, inters = [], []
+        for est in self.estimators_:
+            c = getattr(est, "coef_", None)
+            b = getattr(est, "intercept_", None)
+            if c is None or b is None:
+                return None
+            c = np.asarray(c, dtype=np.float32)
+            coefs.append(c[-1] if c.ndim == 2 else c)
+            inters.append(float(np.ravel(b)[-1]))
+        import torch
+
+        dev = torch.device(device)
+        Wt = torch.as_tensor(np.stack(coefs).T.copy(), device=dev)
+        bt = torch.as_tensor(np.asarray(inters, dtype=np.float32),
+                             device=dev)
+        pairs = np.asarray(self.pairs_, dtype=np.int64)
+        idx_i = torch.as_tensor(pairs[:, 0], device=dev)
+        idx_j = torch.as_tensor(pairs[:, 1], device=dev)
+        k = len(self.classes_)
+        sig = not hasattr(self.estimators_[0], "decision_function")
+
+        def fn(X, chunk=1 << 21):
+            X = np.ascontiguousarray(X, dtype=np.float32)
+            outs = []
+            for lo in range(0, len(X), chunk):
+                xb = torch.as_tensor(X[lo: lo + chunk], device=dev)
+                d = xb @ Wt + bt
+                if sig:
+                    d = torch.sigmoid(d) - 0.5
+                n = len(xb)
+                votes = torch.zeros(n, k, device=dev)
+                conf = torch.zeros(n, k, device=dev)
+                pj = (d > 0).to(torch.float32)
+                votes.index_add_(1, idx_j, pj)
+                votes.index_add_(1, idx_i, 1.0 - pj)
+                conf.index_add_(1, idx_j, d)
+                conf.index_add_(1, idx_i, -d)
+                score = votes + conf / (3 * (conf.abs() + 1))
+                outs.append(score.argmax(dim=1).cpu().numpy())
+            return self.classes_[np.concatenate(outs)]
+
+        return fn
+
 
 def _is_sequence_of_sequences(y):
     if sp.issparse(y) or hasattr(y, "shape"):

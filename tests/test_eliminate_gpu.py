@@ -59,3 +59,19 @@ def test_ovr_device_inference_matches_host():
     predp = DistPredictor(ovr, sc=None, method="predict")
     # fp32 GEMM vs float64 host: near-boundary rows may flip
     assert (predp(X) == ovr.predict(X)).mean() > 0.999
+
+
+def test_ovo_device_inference_matches_host():
+    from skdist_amd.distribute.multiclass import DistOneVsOneClassifier
+
+    rng = np.random.default_rng(4)
+    n, f, k = 6000, 24, 5
+    X = rng.standard_normal((n, f)).astype(np.float32)
+    W = rng.standard_normal((k, f))
+    y = (X @ W.T).argmax(axis=1)
+    ovo = DistOneVsOneClassifier(
+        LogisticRegression(epochs=12, random_state=0),
+        sc=Cluster(require_gpu=True))
+    ovo.fit(X, y)
+    pred = DistPredictor(ovo, sc=None, method="predict")
+    assert (pred(X) == ovo.predict(X)).mean() > 0.999
