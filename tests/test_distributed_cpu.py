@@ -27,8 +27,10 @@ def _spmd_worker(rank, world_size, port, path, q):
     os.environ["RANK"] = str(rank)
     os.environ["WORLD_SIZE"] = str(world_size)
     # CPU/gloo plumbing tests even on a GPU box (two ranks cannot share
-    # the single visible device over NCCL)
+    # the single visible device over NCCL); ROCm reads the HIP/ROCR vars
     os.environ["CUDA_VISIBLE_DEVICES"] = ""
+    os.environ["HIP_VISIBLE_DEVICES"] = ""
+    os.environ["ROCR_VISIBLE_DEVICES"] = ""
     try:
         result = _SCENARIOS[path](rank)
         q.put((rank, "ok", result))
