@@ -225,7 +225,7 @@ class ForestBuilder:
     def __init__(self, ds, criterion, max_depth=None, min_samples_split=2,
                  min_samples_leaf=1, min_impurity_decrease=0.0,
                  max_features=None, extra_mode=False, bootstrap=True,
-                 tree_batch=32, engine=None):
+                 tree_batch=32, engine=None, subtract=True):
         self.ds = ds
         self.crit = _CRITERIA[criterion]
         self.max_depth = 10**9 if max_depth is None else int(max_depth)
@@ -236,6 +236,7 @@ class ForestBuilder:
         self.extra_mode = bool(extra_mode)
         self.bootstrap = bool(bootstrap)
         self.tree_batch = int(tree_batch)
+        self.subtract = bool(subtract)
         if engine is None:
             engine = "hip" if ds.device.type == "cuda" else "eager"
         if engine == "hip":
@@ -310,14 +311,21 @@ class ForestBuilder:
             self._new_node(rec[t])
         fr_start = np.zeros(TB, dtype=np.int64)
         fr_count = counts.astype(np.int64)
+        fr_parent = np.full(TB, -1, dtype=np.int64)
+        fr_sib_start = np.zeros(TB, dtype=np.int64)
+        fr_sib_count = np.zeros(TB, dtype=np.int64)
+        fr_left = np.zeros(TB, dtype=np.int64)
+        prev_hist = None
         depth = 0
         hist_nodes_cap = max(
             1, self.HIST_BUDGET_BYTES // (f * nbins * S * 4))
 
         while len(fr_tree) and depth <= self.max_depth:
-            dec = self._level_decisions(
+            dec, prev_hist = self._level_decisions(
                 seeds, fr_tree, fr_node, fr_start, fr_count,
-                weights, si_a, hist_nodes_cap)
+                weights, si_a, hist_nodes_cap, prev_hist=prev_hist,
+                fr_parent=fr_parent, fr_sib_start=fr_sib_start,
+                fr_sib_count=fr_sib_count, fr_left=fr_left)
             (bfeat, bbin, bgain, bimp, bwl, pstats, lstats) = dec
 
             # host decisions: split or leaf
@@ -338,6 +346,7 @@ class ForestBuilder:
             part_idx = np.flatnonzero(ok)
             nt_tree, nt_node = [], []
             nt_start, nt_count = [], []
+            nt_parent, nt_sstart, nt_scount, nt_left = [], [], [], []
             for i in np.flatnonzero(~ok):
                 t = fr_tree[i]
                 self._make_leaf(rec[t], int(fr_node[i]), pstats[i])
@@ -366,10 +375,12 @@ class ForestBuilder:
                     wr = rs.sum() if ds.is_cls else rs[0]
                     nrows_l = int(nl[k2])
                     nrows_r = int(fr_count[i] - nl[k2])
-                    for (cid, cstats, cw, cstart, ccount) in (
-                        (lid, ls, wl, int(fr_start[i]), nrows_l),
-                        (rid, rs, wr,
-                         int(fr_start[i]) + nrows_l, nrows_r),
+                    st_l = int(fr_start[i])
+                    st_r = st_l + nrows_l
+                    for (cid, cstats, cw, cstart, ccount, sstart,
+                         scount, isl) in (
+                        (lid, ls, wl, st_l, nrows_l, st_r, nrows_r, 1),
+                        (rid, rs, wr, st_r, nrows_r, st_l, nrows_l, 0),
                     ):
                         grow = (
                             depth + 1 < self.max_depth
@@ -382,6 +393,10 @@ class ForestBuilder:
                             nt_node.append(cid)
                             nt_start.append(cstart)
                             nt_count.append(ccount)
+                            nt_parent.append(int(i))
+                            nt_sstart.append(sstart)
+                            nt_scount.append(scount)
+                            nt_left.append(isl)
                         else:
                             self._make_leaf(r, cid, cstats)
                 si_a, si_b = si_b, si_a
@@ -389,6 +404,10 @@ class ForestBuilder:
             fr_node = np.asarray(nt_node, dtype=np.int64)
             fr_start = np.asarray(nt_start, dtype=np.int64)
             fr_count = np.asarray(nt_count, dtype=np.int64)
+            fr_parent = np.asarray(nt_parent, dtype=np.int64)
+            fr_sib_start = np.asarray(nt_sstart, dtype=np.int64)
+            fr_sib_count = np.asarray(nt_scount, dtype=np.int64)
+            fr_left = np.asarray(nt_left, dtype=np.int64)
             depth += 1
 
         return [self._assemble(rec[t]) for t in range(TB)]
@@ -460,15 +479,105 @@ class ForestBuilder:
     # level primitives: hist + split (sliced to the hist budget)
     # -------------------------------------------------------------- #
     def _level_decisions(self, seeds, fr_tree, fr_node, fr_start, fr_count,
-                         weights, si, cap):
+                         weights, si, cap, prev_hist=None, fr_parent=None,
+                         fr_sib_start=None, fr_sib_count=None,
+                         fr_left=None):
+        """Per-frontier-node split decisions.
+
+        Returns (7-tuple of decision arrays, hist_or_None).  When the
+        previous level's histogram tensor is available (``prev_hist``,
+        slot = previous frontier index) the HIP path uses the LightGBM
+        subtraction trick: only the smaller child of each split parent is
+        histogrammed; its sibling's histogram is ``parent − child``
+        (exact for classification — integer-valued f32).  The returned
+        hist tensor feeds the NEXT level's subtraction.
+        """
         NF = len(fr_tree)
+        if (
+            self.engine == "hip"
+            and self.subtract
+            and prev_hist is not None
+            and fr_parent is not None
+            and len(fr_parent) == NF
+            and NF > 0
+            and (fr_parent >= 0).all()
+        ):
+            out = self._subtract_level(
+                seeds, fr_tree, fr_node, fr_start, fr_count, weights, si,
+                cap, prev_hist, fr_parent, fr_sib_start, fr_sib_count,
+                fr_left)
+            if out is not None:
+                return out
+        if self.engine == "hip" and 0 < NF <= cap:
+            dec, hist = self._hist_and_split(
+                seeds, fr_tree, fr_node, fr_start, fr_count, weights, si,
+                return_hist=True)
+            return dec, hist
         outs = []
         for s0 in range(0, NF, cap):
             sl = slice(s0, min(NF, s0 + cap))
             outs.append(self._hist_and_split(
                 seeds, fr_tree[sl], fr_node[sl], fr_start[sl],
                 fr_count[sl], weights, si))
-        return tuple(np.concatenate(parts) for parts in zip(*outs))
+        return (
+            tuple(np.concatenate(parts) for parts in zip(*outs)),
+            None,
+        )
+
+    def _subtract_level(self, seeds, fr_tree, fr_node, fr_start, fr_count,
+                        weights, si, cap, prev_hist, fr_parent,
+                        fr_sib_start, fr_sib_count, fr_left):
+        """One level with sibling-subtraction; None -> caller falls back."""
+        ds = self.ds
+        NF = len(fr_tree)
+        f, nbins, S = ds.f, ds.nbins, ds.S
+        dev = ds.device
+        direct = (fr_count < fr_sib_count) | (
+            (fr_count == fr_sib_count) & (fr_left == 1))
+        # sibling hist source per derived node: the co-frontier sibling
+        # (same parent slot) when present, else an extra slot for the
+        # leaf sibling's rows
+        by_parent = {}
+        for s in range(NF):
+            by_parent.setdefault(int(fr_parent[s]), []).append(s)
+        extras = []          # (tree_slot, start, count)
+        sib_src = np.full(NF, -1, dtype=np.int64)
+        for slots in by_parent.values():
+            if len(slots) == 2:
+                a, b = slots
+                d = b if direct[a] else a
+                sib_src[d] = a if direct[a] else b
+            else:
+                (s,) = slots
+                if not direct[s]:
+                    extras.append((int(fr_tree[s]), int(fr_sib_start[s]),
+                                   int(fr_sib_count[s])))
+                    sib_src[s] = NF + len(extras) - 1
+        total = NF + len(extras)
+        if total > cap:
+            return None
+
+        hist = torch.zeros(total, f, nbins, S, dtype=torch.float32,
+                           device=dev)
+        d_slots = np.flatnonzero(~direct)
+        dir_slots = np.flatnonzero(direct)
+        entries = [
+            (int(s), int(fr_tree[s]), int(fr_start[s]), int(fr_count[s]))
+            for s in dir_slots
+        ] + [
+            (NF + k, t, st, ct) for k, (t, st, ct) in enumerate(extras)
+        ]
+        self._run_hist_kernel(hist, entries, weights, si)
+        if len(d_slots):
+            d_idx = torch.as_tensor(d_slots, device=dev)
+            p_idx = torch.as_tensor(
+                fr_parent[d_slots].astype(np.int64), device=dev)
+            s_idx = torch.as_tensor(sib_src[d_slots], device=dev)
+            hist[d_idx] = prev_hist.index_select(0, p_idx) - \
+                hist.index_select(0, s_idx)
+        dec = self._run_split_kernel(
+            hist, self._node_seeds(seeds, fr_tree, fr_node), NF)
+        return dec, hist
 
     def _node_seeds(self, seeds, fr_tree, fr_node):
         s = np.asarray([seeds[t] for t in fr_tree], dtype=np.int64)
@@ -488,8 +597,51 @@ class ForestBuilder:
                              min(CH, cnt - off)))
         return np.asarray(rows, dtype=np.int32).reshape(-1, 4)
 
+    def _run_hist_kernel(self, hist, entries, weights, si):
+        """Histogram the listed (slot, tree, start, count) segments into
+        ``hist`` (zero-initialized [n_slots, f, nbins, S])."""
+        ds = self.ds
+        dev = ds.device
+        if not entries:
+            return
+        slot, tree, start, count = (np.asarray(a) for a in zip(*entries))
+        chunks_np = self._chunk_table(tree, start, count, slot)
+        chunks = torch.as_tensor(chunks_np, device=dev)
+        self._ext.tree_hist(
+            ds.codes, ds.y_int if ds.is_cls else torch.empty(
+                0, dtype=torch.int32, device=dev),
+            ds.y_f if not ds.is_cls else torch.empty(
+                0, dtype=torch.float32, device=dev),
+            weights, si, chunks, hist, ds.n, ds.f, ds.nbins, ds.S,
+            int(ds.is_cls), self.fg)
+
+    def _run_split_kernel(self, hist, node_seed, NF):
+        ds = self.ds
+        dev = ds.device
+        seed_t = torch.as_tensor(node_seed.astype(np.int32), device=dev)
+        out_feat = torch.empty(NF, dtype=torch.int32, device=dev)
+        out_bin = torch.empty(NF, dtype=torch.int32, device=dev)
+        out_wl = torch.empty(NF, dtype=torch.float32, device=dev)
+        out_gain = torch.empty(NF, dtype=torch.float32, device=dev)
+        out_imp = torch.empty(NF, dtype=torch.float32, device=dev)
+        out_stats = torch.empty(NF, ds.S, dtype=torch.float32, device=dev)
+        out_lstats = torch.empty(NF, ds.S, dtype=torch.float32,
+                                 device=dev)
+        self._ext.tree_split(
+            hist, seed_t, ds.f, ds.nbins, ds.S, int(ds.is_cls), self.crit,
+            self.m_features, int(self.extra_mode), float(self.msl),
+            out_feat, out_bin, out_wl, out_gain, out_imp, out_stats,
+            out_lstats)
+        return (out_feat.cpu().numpy().astype(np.int64),
+                out_bin.cpu().numpy().astype(np.int64),
+                out_gain.cpu().numpy().astype(np.float64),
+                out_imp.cpu().numpy().astype(np.float64),
+                out_wl.cpu().numpy().astype(np.float64),
+                out_stats.cpu().numpy().astype(np.float64),
+                out_lstats.cpu().numpy().astype(np.float64))
+
     def _hist_and_split(self, seeds, fr_tree, fr_node, fr_start, fr_count,
-                        weights, si):
+                        weights, si, return_hist=False):
         ds = self.ds
         NF = len(fr_tree)
         f, nbins, S = ds.f, ds.nbins, ds.S
@@ -497,40 +649,17 @@ class ForestBuilder:
         node_seed = self._node_seeds(seeds, fr_tree, fr_node)
 
         if self.engine == "hip":
-            chunks_np = self._chunk_table(
-                fr_tree, fr_start, fr_count, np.arange(NF))
-            chunks = torch.as_tensor(chunks_np, device=dev)
             hist = torch.zeros(NF, f, nbins, S, dtype=torch.float32,
                                device=dev)
-            self._ext.tree_hist(
-                ds.codes, ds.y_int if ds.is_cls else torch.empty(
-                    0, dtype=torch.int32, device=dev),
-                ds.y_f if not ds.is_cls else torch.empty(
-                    0, dtype=torch.float32, device=dev),
-                weights, si, chunks, hist, ds.n, f, nbins, S,
-                int(ds.is_cls), self.fg)
-            seed_t = torch.as_tensor(
-                node_seed.astype(np.int32), device=dev)
-            out_feat = torch.empty(NF, dtype=torch.int32, device=dev)
-            out_bin = torch.empty(NF, dtype=torch.int32, device=dev)
-            out_wl = torch.empty(NF, dtype=torch.float32, device=dev)
-            out_gain = torch.empty(NF, dtype=torch.float32, device=dev)
-            out_imp = torch.empty(NF, dtype=torch.float32, device=dev)
-            out_stats = torch.empty(NF, S, dtype=torch.float32, device=dev)
-            out_lstats = torch.empty(NF, S, dtype=torch.float32,
-                                     device=dev)
-            self._ext.tree_split(
-                hist, seed_t, f, nbins, S, int(ds.is_cls), self.crit,
-                self.m_features, int(self.extra_mode), float(self.msl),
-                out_feat, out_bin, out_wl, out_gain, out_imp, out_stats,
-                out_lstats)
-            return (out_feat.cpu().numpy().astype(np.int64),
-                    out_bin.cpu().numpy().astype(np.int64),
-                    out_gain.cpu().numpy().astype(np.float64),
-                    out_imp.cpu().numpy().astype(np.float64),
-                    out_wl.cpu().numpy().astype(np.float64),
-                    out_stats.cpu().numpy().astype(np.float64),
-                    out_lstats.cpu().numpy().astype(np.float64))
+            entries = [
+                (s, int(fr_tree[s]), int(fr_start[s]), int(fr_count[s]))
+                for s in range(NF)
+            ]
+            self._run_hist_kernel(hist, entries, weights, si)
+            dec = self._run_split_kernel(hist, node_seed, NF)
+            if return_hist:
+                return dec, hist
+            return dec
 
         return self._hist_and_split_eager(
             fr_tree, fr_start, fr_count, weights, si, node_seed)

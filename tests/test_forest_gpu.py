@@ -179,3 +179,20 @@ def test_sklearn_forest_and_gbt_on_device():
     pred = DistPredictor(gbt, sc=None, method="predict_proba")
     np.testing.assert_allclose(
         pred(X), gbt.predict_proba(X), atol=1e-4)
+
+
+def test_subtraction_trick_identical_trees():
+    """Sibling-subtraction histograms are exact for classification: the
+    built trees must match the direct-histogram path bit-for-bit."""
+    X, y = _cls_data(5000, 12, seed=11)
+    ds = BinnedDataset(X, y, "cuda", is_cls=True)
+    kw = dict(max_depth=9, max_features="sqrt", bootstrap=True,
+              tree_batch=4)
+    a = ForestBuilder(ds, "gini", subtract=True, **kw).build([5, 21])
+    b = ForestBuilder(ds, "gini", subtract=False, **kw).build([5, 21])
+    for ta, tb in zip(a, b):
+        assert ta.node_count == tb.node_count
+        np.testing.assert_array_equal(ta.feature, tb.feature)
+        np.testing.assert_array_equal(ta.threshold, tb.threshold)
+        np.testing.assert_array_equal(ta.left, tb.left)
+        np.testing.assert_allclose(ta.value, tb.value, atol=1e-6)
