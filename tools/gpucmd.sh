@@ -1,13 +1,24 @@
 export TMPDIR=/tmp
 mkdir -p /root/repo/gpurun_out
 cd /root/repo
-echo "== build check (extension fresh) =="
-python -c "from skdist_amd.ops.build import extension_is_stale; print('stale:', extension_is_stale())"
-echo "== full gpu suite =="
-timeout 900 python -m pytest tests -m gpu -q 2>&1 | tail -2
-echo "== graft smoke =="
-timeout 300 python -c "import __graft_entry__ as g; g.smoke()" 2>&1 | tail -1
-echo "== bench =="
-timeout 700 python bench.py --steps 3 --warmup 1 2>/dev/null | tail -1 | tee gpurun_out/bench_final.json | python3 -c "import json,sys; d=json.load(sys.stdin); print(round(d['ms_per_step'],1),'ms/step', round(d['value']),'fits/s', 'vs_baseline', round(d['vs_baseline'],1))"
-echo "== non-gpu tests on gpu box (should all pass/skip cleanly) =="
-timeout 900 python -m pytest tests -m "not gpu" -q -x 2>&1 | tail -1
+echo "== forest gpu tests (incl. subtraction parity) =="
+timeout 900 python -m pytest tests/test_forest_gpu.py -x -q 2>&1 | tail -2
+echo "== forest perf: subtract vs direct, 1M x 64 depth 12, 64 trees =="
+PYTHONPATH=/root/repo timeout 900 python - <<'PY' 2>&1 | tail -4
+import numpy as np, time, torch
+from skdist_amd.models.forest import BinnedDataset, ForestBuilder
+rng = np.random.default_rng(0)
+n, f = 1_000_000, 64
+X = rng.standard_normal((n, f)).astype(np.float32)
+y = ((X @ rng.standard_normal(f)) > 0).astype(np.int64)
+ds = BinnedDataset(X, y, "cuda", is_cls=True)
+for sub in (False, True, False, True):
+    b = ForestBuilder(ds, "gini", max_depth=12, max_features="sqrt", bootstrap=True, tree_batch=32, subtract=sub)
+    b.build([0])
+    torch.cuda.synchronize(); t0 = time.time()
+    trees = b.build(list(range(64)))
+    torch.cuda.synchronize(); dt = time.time()-t0
+    print(f"subtract={sub}: {dt:.2f}s = {64/dt:.2f} trees/s, nodes {np.mean([t.node_count for t in trees]):.0f}")
+PY
+echo "== config3 probe with subtraction =="
+PYTHONPATH=/root/repo timeout 600 python tools/config_probes.py forest 2>&1 | grep '^{'
