@@ -212,6 +212,111 @@ class BinnedDataset:
 # the level-synchronous builder
 # --------------------------------------------------------------------- #
 
+
+class _NodeStore:
+    """Flat growing node table for one tree batch (host, numpy).
+
+    Nodes of every tree in the batch share one store; per-tree node ids
+    (``nid``) are assigned densely in creation order so ``assemble``
+    reproduces exactly the per-tree layout the scalar builder used.
+    All level bookkeeping is vectorized over this store — the builder's
+    wall time was host-bound before (kernels ~125 ms vs ~1.2 s wall for
+    a 32-tree batch), dominated by per-node Python loops.
+    """
+
+    def __init__(self, TB, s_out):
+        self.s_out = s_out
+        self.cap = 1024
+        self.n = 0
+        self.tree = np.empty(self.cap, dtype=np.int64)
+        self.nid = np.empty(self.cap, dtype=np.int64)
+        self.feature = np.empty(self.cap, dtype=np.int64)
+        self.bin = np.empty(self.cap, dtype=np.int64)
+        self.left = np.empty(self.cap, dtype=np.int64)
+        self.right = np.empty(self.cap, dtype=np.int64)
+        self.leafrow = np.empty(self.cap, dtype=np.int64)
+        self.tree_counts = np.zeros(TB, dtype=np.int64)
+        self.leaf_blocks = []
+        self.n_leaves = 0
+
+    def _ensure(self, extra):
+        need = self.n + extra
+        if need <= self.cap:
+            return
+        while self.cap < need:
+            self.cap *= 2
+        for name in ("tree", "nid", "feature", "bin", "left", "right",
+                     "leafrow"):
+            old = getattr(self, name)
+            new = np.empty(self.cap, dtype=np.int64)
+            new[: self.n] = old[: self.n]
+            setattr(self, name, new)
+
+    def append(self, tree, nid):
+        m = len(tree)
+        self._ensure(m)
+        lo = self.n
+        self.n += m
+        sl = slice(lo, self.n)
+        self.tree[sl] = tree
+        self.nid[sl] = nid
+        self.feature[sl] = -1
+        self.bin[sl] = -1
+        self.left[sl] = -1
+        self.right[sl] = -1
+        self.leafrow[sl] = -1
+        return np.arange(lo, self.n)
+
+    def make_leaves(self, rows, values):
+        """Mark store rows as leaves with the given [m, s_out] payloads."""
+        m = len(rows)
+        if m == 0:
+            return
+        self.feature[rows] = -1
+        self.leafrow[rows] = self.n_leaves + np.arange(m)
+        self.n_leaves += m
+        self.leaf_blocks.append(
+            np.ascontiguousarray(values, dtype=np.float32))
+
+    def assemble(self, ds, importances):
+        """Per-tree HistTree list (nid-ordered arrays, leaf re-indexed)."""
+        n = self.n
+        vals = (
+            np.concatenate(self.leaf_blocks, axis=0)
+            if self.leaf_blocks
+            else np.zeros((0, self.s_out), dtype=np.float32)
+        )
+        edges = ds.edges_np()
+        order = np.lexsort((self.nid[:n], self.tree[:n]))
+        tree_sorted = self.tree[:n][order]
+        bounds = np.searchsorted(
+            tree_sorted, np.arange(len(self.tree_counts) + 1))
+        out = []
+        for t in range(len(self.tree_counts)):
+            rows = order[bounds[t]: bounds[t + 1]]
+            feature = self.feature[rows].astype(np.int32)
+            bins = self.bin[rows]
+            thr = np.zeros(len(rows), dtype=np.float32)
+            internal = feature >= 0
+            if internal.any():
+                thr[internal] = edges[feature[internal], bins[internal]]
+            left = self.left[rows].astype(np.int32)
+            right = self.right[rows].astype(np.int32)
+            leaf = ~internal
+            lrows = self.leafrow[rows][leaf]
+            value = (
+                vals[lrows] if len(lrows)
+                else np.zeros((1, self.s_out), dtype=np.float32)
+            )
+            left[leaf] = np.arange(leaf.sum())
+            imp = importances[t]
+            ssum = imp.sum()
+            out.append(HistTree(
+                feature, thr, left, right, value, ds.classes_, ds.f,
+                (imp / ssum if ssum > 0 else imp).astype(np.float64)))
+        return out
+
+
 class ForestBuilder:
     """Grows a batch of trees level-by-level against one BinnedDataset.
 
@@ -290,25 +395,20 @@ class ForestBuilder:
     def _build_batch(self, seeds, sample_weight):
         ds = self.ds
         TB = len(seeds)
-        n, f, nbins, S = ds.n, ds.f, ds.nbins, ds.S
-        dev = ds.device
+        f, nbins, S = ds.f, ds.nbins, ds.S
+        s_out = S if ds.is_cls else 1
 
         weights = self.make_weights(seeds, sample_weight)
         si_a, counts = self._initial_sample_idx(weights)
         si_b = torch.empty_like(si_a)
 
-        # per-tree growing node records (host)
-        rec = [
-            {"feature": [], "bin": [], "left": [], "right": [],
-             "leaf_val": [], "importance": np.zeros(f), "root_w": 1.0}
-            for _ in range(TB)
-        ]
+        st = _NodeStore(TB, s_out)
+        importances = np.zeros((TB, f))
+        root_w = np.ones(TB)
 
-        # frontier: numpy arrays (tree_slot, node_id, seg_start, seg_count)
         fr_tree = np.arange(TB, dtype=np.int64)
-        fr_node = np.zeros(TB, dtype=np.int64)
-        for t in range(TB):
-            self._new_node(rec[t])
+        fr_grow = st.append(fr_tree, np.zeros(TB, dtype=np.int64))
+        st.tree_counts[:] = 1
         fr_start = np.zeros(TB, dtype=np.int64)
         fr_count = counts.astype(np.int64)
         fr_parent = np.full(TB, -1, dtype=np.int64)
@@ -322,97 +422,121 @@ class ForestBuilder:
 
         while len(fr_tree) and depth <= self.max_depth:
             dec, prev_hist = self._level_decisions(
-                seeds, fr_tree, fr_node, fr_start, fr_count,
+                seeds, fr_tree, st.nid[fr_grow], fr_start, fr_count,
                 weights, si_a, hist_nodes_cap, prev_hist=prev_hist,
                 fr_parent=fr_parent, fr_sib_start=fr_sib_start,
                 fr_sib_count=fr_sib_count, fr_left=fr_left)
             (bfeat, bbin, bgain, bimp, bwl, pstats, lstats) = dec
 
-            # host decisions: split or leaf
-            wp = (pstats.sum(axis=1) if ds.is_cls else pstats[:, 0])
-            root_scale = np.empty(len(fr_tree))
-            for i, t in enumerate(fr_tree):
-                if fr_node[i] == 0:
-                    rec[t]["root_w"] = max(float(wp[i]), 1.0)
-                root_scale[i] = rec[t]["root_w"]
+            wp = pstats.sum(axis=1) if ds.is_cls else pstats[:, 0]
+            if depth == 0:
+                root_w[fr_tree] = np.maximum(wp, 1.0)
             ok = (
                 (bfeat >= 0)
                 & (depth < self.max_depth)
                 & (wp >= self.mss)
                 & (bgain > 0)
-                & ((wp / root_scale) * bgain >= self.mid - 1e-12)
+                & ((wp / root_w[fr_tree]) * bgain >= self.mid - 1e-12)
             )
 
             part_idx = np.flatnonzero(ok)
-            nt_tree, nt_node = [], []
-            nt_start, nt_count = [], []
-            nt_parent, nt_sstart, nt_scount, nt_left = [], [], [], []
-            for i in np.flatnonzero(~ok):
-                t = fr_tree[i]
-                self._make_leaf(rec[t], int(fr_node[i]), pstats[i])
-            # child bookkeeping requires partition counts first
-            if len(part_idx):
-                nl = self._partition(
-                    fr_tree[part_idx], fr_start[part_idx],
-                    fr_count[part_idx], bfeat[part_idx], bbin[part_idx],
-                    si_a, si_b)
-                for k2, i in enumerate(part_idx):
-                    t = int(fr_tree[i])
-                    node = int(fr_node[i])
-                    r = rec[t]
-                    jf, jb = int(bfeat[i]), int(bbin[i])
-                    lid = self._new_node(r)
-                    rid = self._new_node(r)
-                    r["feature"][node] = jf
-                    r["bin"][node] = jb
-                    r["left"][node] = lid
-                    r["right"][node] = rid
-                    r["importance"][jf] += (
-                        float(wp[i]) * float(bgain[i]) / r["root_w"])
-                    ls = lstats[i]
-                    rs = pstats[i] - ls
-                    wl = ls.sum() if ds.is_cls else ls[0]
-                    wr = rs.sum() if ds.is_cls else rs[0]
-                    nrows_l = int(nl[k2])
-                    nrows_r = int(fr_count[i] - nl[k2])
-                    st_l = int(fr_start[i])
-                    st_r = st_l + nrows_l
-                    for (cid, cstats, cw, cstart, ccount, sstart,
-                         scount, isl) in (
-                        (lid, ls, wl, st_l, nrows_l, st_r, nrows_r, 1),
-                        (rid, rs, wr, st_r, nrows_r, st_l, nrows_l, 0),
-                    ):
-                        grow = (
-                            depth + 1 < self.max_depth
-                            and cw >= self.mss
-                            and ccount > 1
-                            and self._impurity_np(cstats, cw) > 1e-12
-                        )
-                        if grow:
-                            nt_tree.append(t)
-                            nt_node.append(cid)
-                            nt_start.append(cstart)
-                            nt_count.append(ccount)
-                            nt_parent.append(int(i))
-                            nt_sstart.append(sstart)
-                            nt_scount.append(scount)
-                            nt_left.append(isl)
-                        else:
-                            self._make_leaf(r, cid, cstats)
-                si_a, si_b = si_b, si_a
-            fr_tree = np.asarray(nt_tree, dtype=np.int64)
-            fr_node = np.asarray(nt_node, dtype=np.int64)
-            fr_start = np.asarray(nt_start, dtype=np.int64)
-            fr_count = np.asarray(nt_count, dtype=np.int64)
-            fr_parent = np.asarray(nt_parent, dtype=np.int64)
-            fr_sib_start = np.asarray(nt_sstart, dtype=np.int64)
-            fr_sib_count = np.asarray(nt_scount, dtype=np.int64)
-            fr_left = np.asarray(nt_left, dtype=np.int64)
+            leaf_idx = np.flatnonzero(~ok)
+            st.make_leaves(fr_grow[leaf_idx],
+                           self._leaf_values_vec(pstats[leaf_idx]))
+
+            if not len(part_idx):
+                fr_tree = np.empty(0, dtype=np.int64)
+                depth += 1
+                continue
+
+            nl = self._partition(
+                fr_tree[part_idx], fr_start[part_idx],
+                fr_count[part_idx], bfeat[part_idx], bbin[part_idx],
+                si_a, si_b)
+            si_a, si_b = si_b, si_a
+
+            t_p = fr_tree[part_idx]
+            # per-tree dense child-id assignment in part order
+            order = np.argsort(t_p, kind="stable")
+            ts = t_p[order]
+            starts = np.flatnonzero(np.r_[True, ts[1:] != ts[:-1]])
+            seg_len = np.diff(np.r_[starts, len(ts)])
+            ranks = np.empty(len(t_p), dtype=np.int64)
+            ranks[order] = np.arange(len(ts)) - np.repeat(starts, seg_len)
+            lid = st.tree_counts[t_p] + 2 * ranks
+            rid = lid + 1
+            st.tree_counts += 2 * np.bincount(t_p, minlength=TB)
+
+            gl = st.append(t_p, lid)
+            gr = st.append(t_p, rid)
+            pg = fr_grow[part_idx]
+            st.feature[pg] = bfeat[part_idx]
+            st.bin[pg] = bbin[part_idx]
+            st.left[pg] = lid
+            st.right[pg] = rid
+            np.add.at(
+                importances, (t_p, bfeat[part_idx]),
+                wp[part_idx] * bgain[part_idx] / root_w[t_p])
+
+            ls = lstats[part_idx]
+            rs = pstats[part_idx] - ls
+            if ds.is_cls:
+                wl = ls.sum(axis=1)
+                wr = rs.sum(axis=1)
+            else:
+                wl = ls[:, 0]
+                wr = rs[:, 0]
+            nrows_l = nl.astype(np.int64)
+            nrows_r = fr_count[part_idx] - nrows_l
+            st_l = fr_start[part_idx]
+            st_r = st_l + nrows_l
+
+            can_grow = depth + 1 < self.max_depth
+            grow_l = (can_grow & (wl >= self.mss) & (nrows_l > 1)
+                      & (self._impurity_vec(ls, wl) > 1e-12))
+            grow_r = (can_grow & (wr >= self.mss) & (nrows_r > 1)
+                      & (self._impurity_vec(rs, wr) > 1e-12))
+            st.make_leaves(gl[~grow_l], self._leaf_values_vec(ls[~grow_l]))
+            st.make_leaves(gr[~grow_r], self._leaf_values_vec(rs[~grow_r]))
+
+            fr_tree = np.concatenate([t_p[grow_l], t_p[grow_r]])
+            fr_grow = np.concatenate([gl[grow_l], gr[grow_r]])
+            fr_start = np.concatenate([st_l[grow_l], st_r[grow_r]])
+            fr_count = np.concatenate([nrows_l[grow_l], nrows_r[grow_r]])
+            fr_parent = np.concatenate(
+                [part_idx[grow_l], part_idx[grow_r]])
+            fr_sib_start = np.concatenate([st_r[grow_l], st_l[grow_r]])
+            fr_sib_count = np.concatenate(
+                [nrows_r[grow_l], nrows_l[grow_r]])
+            fr_left = np.concatenate([
+                np.ones(int(grow_l.sum()), dtype=np.int64),
+                np.zeros(int(grow_r.sum()), dtype=np.int64),
+            ])
             depth += 1
 
-        return [self._assemble(rec[t]) for t in range(TB)]
+        return st.assemble(ds, importances)
 
-    # -------------------------------------------------------------- #
+    def _leaf_values_vec(self, stats):
+        """[m, S] stats -> [m, s_out] leaf payloads (vectorized)."""
+        stats = np.atleast_2d(stats)
+        if self.ds.is_cls:
+            tot = np.clip(stats.sum(axis=1, keepdims=True), 1e-30, None)
+            return stats / tot
+        return (stats[:, 1] / np.clip(stats[:, 0], 1e-30, None))[:, None]
+
+    def _impurity_vec(self, stats, w):
+        """Vectorized impurity over [m, S] stats rows."""
+        w = np.clip(np.asarray(w, dtype=np.float64), 1e-30, None)
+        if self.ds.is_cls:
+            if self.crit == CRIT_GINI:
+                return 1.0 - (stats ** 2).sum(axis=1) / (w * w)
+            p = stats / w[:, None]
+            with np.errstate(divide="ignore", invalid="ignore"):
+                e = np.where(p > 0, p * np.log2(p, where=p > 0), 0.0)
+            return -e.sum(axis=1)
+        mean = stats[:, 1] / w
+        return np.maximum(stats[:, 2] / w - mean * mean, 0.0)
+
     def _initial_sample_idx(self, weights):
         """Pack rows with weight > 0 to the front of each tree's row."""
         ds = self.ds
@@ -425,23 +549,6 @@ class ForestBuilder:
             si[t, : len(nz)] = nz.to(torch.int32)
         return si, counts
 
-    def _new_node(self, r):
-        r["feature"].append(-1)
-        r["bin"].append(-1)
-        r["left"].append(-1)
-        r["right"].append(-1)
-        return len(r["feature"]) - 1
-
-    def _make_leaf(self, r, node, stats):
-        r["feature"][node] = -1
-        r["left"][node] = len(r["leaf_val"])
-        if self.ds.is_cls:
-            tot = stats.sum()
-            r["leaf_val"].append(stats / max(tot, 1e-30))
-        else:
-            r["leaf_val"].append(
-                np.array([stats[1] / max(stats[0], 1e-30)]))
-
     def _impurity_np(self, stats, w):
         if w <= 0:
             return 0.0
@@ -452,28 +559,6 @@ class ForestBuilder:
             return float(-(p * np.log2(p)).sum())
         mean = stats[1] / w
         return max(float(stats[2] / w - mean * mean), 0.0)
-
-    def _assemble(self, r):
-        ds = self.ds
-        feature = np.asarray(r["feature"], dtype=np.int32)
-        bins = np.asarray(r["bin"], dtype=np.int32)
-        left = np.asarray(r["left"], dtype=np.int32)
-        right = np.asarray(r["right"], dtype=np.int32)
-        thr = np.zeros(len(feature), dtype=np.float32)
-        internal = feature >= 0
-        if internal.any():
-            edges = ds.edges_np()
-            thr[internal] = edges[feature[internal], bins[internal]]
-        if r["leaf_val"]:
-            value = np.stack(r["leaf_val"]).astype(np.float32)
-        else:  # degenerate: no samples at all
-            vs = ds.S if ds.is_cls else 1
-            value = np.zeros((1, vs), dtype=np.float32)
-        imp = r["importance"]
-        s = imp.sum()
-        return HistTree(feature, thr, left, right, value,
-                        ds.classes_, ds.f,
-                        (imp / s if s > 0 else imp).astype(np.float64))
 
     # -------------------------------------------------------------- #
     # level primitives: hist + split (sliced to the hist budget)
@@ -537,22 +622,20 @@ class ForestBuilder:
         # sibling hist source per derived node: the co-frontier sibling
         # (same parent slot) when present, else an extra slot for the
         # leaf sibling's rows
-        by_parent = {}
-        for s in range(NF):
-            by_parent.setdefault(int(fr_parent[s]), []).append(s)
-        extras = []          # (tree_slot, start, count)
         sib_src = np.full(NF, -1, dtype=np.int64)
-        for slots in by_parent.values():
-            if len(slots) == 2:
-                a, b = slots
-                d = b if direct[a] else a
-                sib_src[d] = a if direct[a] else b
-            else:
-                (s,) = slots
-                if not direct[s]:
-                    extras.append((int(fr_tree[s]), int(fr_sib_start[s]),
-                                   int(fr_sib_count[s])))
-                    sib_src[s] = NF + len(extras) - 1
+        order = np.argsort(fr_parent, kind="stable")
+        ps = fr_parent[order]
+        starts = np.flatnonzero(np.r_[True, ps[1:] != ps[:-1]])
+        seg = np.diff(np.r_[starts, NF])
+        pair_at = starts[seg == 2]
+        a = order[pair_at]
+        b = order[pair_at + 1]
+        da = direct[a]
+        sib_src[np.where(da, b, a)] = np.where(da, a, b)
+        singles = order[starts[seg == 1]]
+        need = singles[~direct[singles]]  # derive from a leaf sibling
+        sib_src[need] = NF + np.arange(len(need))
+        extras = need
         total = NF + len(extras)
         if total > cap:
             return None
@@ -561,13 +644,14 @@ class ForestBuilder:
                            device=dev)
         d_slots = np.flatnonzero(~direct)
         dir_slots = np.flatnonzero(direct)
-        entries = [
-            (int(s), int(fr_tree[s]), int(fr_start[s]), int(fr_count[s]))
-            for s in dir_slots
-        ] + [
-            (NF + k, t, st, ct) for k, (t, st, ct) in enumerate(extras)
-        ]
-        self._run_hist_kernel(hist, entries, weights, si)
+        h_slot = np.concatenate([dir_slots, NF + np.arange(len(extras))])
+        h_tree = np.concatenate([fr_tree[dir_slots], fr_tree[extras]])
+        h_start = np.concatenate(
+            [fr_start[dir_slots], fr_sib_start[extras]])
+        h_count = np.concatenate(
+            [fr_count[dir_slots], fr_sib_count[extras]])
+        self._run_hist_kernel(hist, h_slot, h_tree, h_start, h_count,
+                              weights, si)
         if len(d_slots):
             d_idx = torch.as_tensor(d_slots, device=dev)
             p_idx = torch.as_tensor(
@@ -586,25 +670,30 @@ class ForestBuilder:
         return _wang_hash(x.astype(_U32)).astype(np.uint32)
 
     def _chunk_table(self, fr_tree, fr_start, fr_count, slot_ids):
-        """[n_chunks, 4] = {slot, tree_slot, row_start, row_count}."""
+        """[n_chunks, 4] = {slot, tree_slot, row_start, row_count}
+        (vectorized; a node always yields >= 1 chunk)."""
         CH = self.CHUNK_ROWS
-        rows = []
-        for k in range(len(fr_tree)):
-            cnt = int(fr_count[k])
-            st = int(fr_start[k])
-            for off in range(0, max(cnt, 1), CH):
-                rows.append((int(slot_ids[k]), int(fr_tree[k]), st + off,
-                             min(CH, cnt - off)))
-        return np.asarray(rows, dtype=np.int32).reshape(-1, 4)
+        fr_count = np.asarray(fr_count, dtype=np.int64)
+        nch = np.maximum(1, (fr_count + CH - 1) // CH)
+        rep = np.repeat(np.arange(len(fr_tree)), nch)
+        firsts = np.cumsum(nch) - nch
+        within = np.arange(len(rep)) - np.repeat(firsts, nch)
+        off = within * CH
+        out = np.empty((len(rep), 4), dtype=np.int32)
+        out[:, 0] = np.asarray(slot_ids)[rep]
+        out[:, 1] = np.asarray(fr_tree)[rep]
+        out[:, 2] = np.asarray(fr_start)[rep] + off
+        out[:, 3] = np.minimum(CH, fr_count[rep] - off)
+        return out
 
-    def _run_hist_kernel(self, hist, entries, weights, si):
-        """Histogram the listed (slot, tree, start, count) segments into
-        ``hist`` (zero-initialized [n_slots, f, nbins, S])."""
+    def _run_hist_kernel(self, hist, slot, tree, start, count, weights,
+                         si):
+        """Histogram the given segments (parallel arrays) into ``hist``
+        (zero-initialized [n_slots, f, nbins, S])."""
         ds = self.ds
         dev = ds.device
-        if not entries:
+        if not len(slot):
             return
-        slot, tree, start, count = (np.asarray(a) for a in zip(*entries))
         chunks_np = self._chunk_table(tree, start, count, slot)
         chunks = torch.as_tensor(chunks_np, device=dev)
         self._ext.tree_hist(
@@ -618,27 +707,33 @@ class ForestBuilder:
     def _run_split_kernel(self, hist, node_seed, NF):
         ds = self.ds
         dev = ds.device
+        S = ds.S
         seed_t = torch.as_tensor(node_seed.astype(np.int32), device=dev)
-        out_feat = torch.empty(NF, dtype=torch.int32, device=dev)
-        out_bin = torch.empty(NF, dtype=torch.int32, device=dev)
-        out_wl = torch.empty(NF, dtype=torch.float32, device=dev)
-        out_gain = torch.empty(NF, dtype=torch.float32, device=dev)
-        out_imp = torch.empty(NF, dtype=torch.float32, device=dev)
-        out_stats = torch.empty(NF, ds.S, dtype=torch.float32, device=dev)
-        out_lstats = torch.empty(NF, ds.S, dtype=torch.float32,
-                                 device=dev)
+        # outputs packed into two buffers -> two D2H transfers per level
+        ibuf = torch.empty(2 * NF, dtype=torch.int32, device=dev)
+        fbuf = torch.empty((3 + 2 * S) * NF, dtype=torch.float32,
+                           device=dev)
+        out_feat = ibuf[:NF]
+        out_bin = ibuf[NF:]
+        out_wl = fbuf[:NF]
+        out_gain = fbuf[NF: 2 * NF]
+        out_imp = fbuf[2 * NF: 3 * NF]
+        out_stats = fbuf[3 * NF: (3 + S) * NF].view(NF, S)
+        out_lstats = fbuf[(3 + S) * NF:].view(NF, S)
         self._ext.tree_split(
-            hist, seed_t, ds.f, ds.nbins, ds.S, int(ds.is_cls), self.crit,
+            hist, seed_t, ds.f, ds.nbins, S, int(ds.is_cls), self.crit,
             self.m_features, int(self.extra_mode), float(self.msl),
             out_feat, out_bin, out_wl, out_gain, out_imp, out_stats,
             out_lstats)
-        return (out_feat.cpu().numpy().astype(np.int64),
-                out_bin.cpu().numpy().astype(np.int64),
-                out_gain.cpu().numpy().astype(np.float64),
-                out_imp.cpu().numpy().astype(np.float64),
-                out_wl.cpu().numpy().astype(np.float64),
-                out_stats.cpu().numpy().astype(np.float64),
-                out_lstats.cpu().numpy().astype(np.float64))
+        ih = ibuf.cpu().numpy()
+        fh = fbuf.cpu().numpy().astype(np.float64)
+        return (ih[:NF].astype(np.int64),
+                ih[NF:].astype(np.int64),
+                fh[NF: 2 * NF],
+                fh[2 * NF: 3 * NF],
+                fh[:NF],
+                fh[3 * NF: (3 + S) * NF].reshape(NF, S),
+                fh[(3 + S) * NF:].reshape(NF, S))
 
     def _hist_and_split(self, seeds, fr_tree, fr_node, fr_start, fr_count,
                         weights, si, return_hist=False):
@@ -651,11 +746,8 @@ class ForestBuilder:
         if self.engine == "hip":
             hist = torch.zeros(NF, f, nbins, S, dtype=torch.float32,
                                device=dev)
-            entries = [
-                (s, int(fr_tree[s]), int(fr_start[s]), int(fr_count[s]))
-                for s in range(NF)
-            ]
-            self._run_hist_kernel(hist, entries, weights, si)
+            self._run_hist_kernel(hist, np.arange(NF), fr_tree, fr_start,
+                                  fr_count, weights, si)
             dec = self._run_split_kernel(hist, node_seed, NF)
             if return_hist:
                 return dec, hist
@@ -820,20 +912,21 @@ class ForestBuilder:
         counts_np = counts.cpu().numpy().astype(np.int64)
 
         # per-node prefix over its chunks → absolute left/right bases
+        # (vectorized: the chunk table is ordered per node, so group
+        # prefixes are global cumsums minus each group's start value)
         slot = chunks_np[:, 0].astype(np.int64)
         rows_np = chunks_np[:, 3].astype(np.int64)
-        nl = np.zeros(NP, dtype=np.int64)
-        np.add.at(nl, slot, counts_np)
-        lbase = np.empty(nch, dtype=np.int64)
-        rbase = np.empty(nch, dtype=np.int64)
-        run_l = np.zeros(NP, dtype=np.int64)
-        run_r = np.zeros(NP, dtype=np.int64)
-        for c in range(nch):  # chunk table is ordered per node
-            k = slot[c]
-            lbase[c] = p_start[k] + run_l[k]
-            rbase[c] = p_start[k] + nl[k] + run_r[k]
-            run_l[k] += counts_np[c]
-            run_r[k] += rows_np[c] - counts_np[c]
+        nl = np.bincount(slot, weights=counts_np,
+                         minlength=NP).astype(np.int64)
+        starts = np.flatnonzero(np.r_[True, slot[1:] != slot[:-1]])
+        seg_len = np.diff(np.r_[starts, nch])
+        ex_l = np.cumsum(counts_np) - counts_np
+        run_l = ex_l - np.repeat(ex_l[starts], seg_len)
+        rminusc = rows_np - counts_np
+        ex_r = np.cumsum(rminusc) - rminusc
+        run_r = ex_r - np.repeat(ex_r[starts], seg_len)
+        lbase = p_start[slot] + run_l
+        rbase = p_start[slot] + nl[slot] + run_r
         lb = torch.as_tensor(lbase.astype(np.int32), device=dev)
         rb = torch.as_tensor(rbase.astype(np.int32), device=dev)
         self._ext.part_scatter(ds.codes, si_in, chunks, feat_t, bin_t,
