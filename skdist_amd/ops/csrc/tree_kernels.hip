@@ -460,7 +460,10 @@ extern "C" hipError_t skdist_tree_split(
     float min_leaf_w, void* out_feat, void* out_bin, void* out_wl,
     void* out_gain, void* out_imp, void* out_stats, void* out_lstats,
     hipStream_t stream) {
-    hipLaunchKernelGGL(k_tree_split, dim3(n_frontier), dim3(256), 0,
+    // per-node parallelism is one thread per feature: size the block to
+    // f so low-f datasets don't idle 3/4 of each workgroup
+    const int threads = f >= 192 ? 256 : (f >= 96 ? 128 : 64);
+    hipLaunchKernelGGL(k_tree_split, dim3(n_frontier), dim3(threads), 0,
                        stream, (const float*)hist,
                        (const unsigned*)node_seed, f, nbins, S, is_cls,
                        crit, m_features, extra_mode, min_leaf_w,
