@@ -183,6 +183,26 @@ def _scenario_predictor(rank):
     return {"match": bool(np.allclose(out, ref)), "n": len(out)}
 
 
+def _scenario_encoder(rank):
+    import pandas as pd
+
+    from skdist_amd import Cluster
+    from skdist_amd.distribute.encoder import Encoderizer
+
+    sc = Cluster()
+    df = None
+    if rank == 0:
+        df = pd.DataFrame({
+            "txt": [f"doc {i} topic {i % 3}" for i in range(80)],
+            "num": np.arange(80, dtype=float),
+            "cat": [f"c{i % 4}" for i in range(80)],
+        })
+    df = sc.sync_host_data(df)
+    enc = Encoderizer(size="small", sc=sc)
+    T = enc.fit_transform(df)
+    return {"shape": [int(v) for v in T.shape]}
+
+
 _SCENARIOS = {
     "batched": _scenario_batched,
     "generic": _scenario_generic,
@@ -190,6 +210,7 @@ _SCENARIOS = {
     "ovr": _scenario_ovr,
     "eliminate": _scenario_eliminate,
     "predictor": _scenario_predictor,
+    "encoder": _scenario_encoder,
 }
 
 
@@ -261,3 +282,10 @@ def test_spmd_predictor_gloo():
     outs = _run_spmd("predictor")
     assert outs[0]["match"] and outs[1]["match"]
     assert outs[0]["n"] == 2000
+
+
+@pytest.mark.timeout(300)
+def test_spmd_encoder_gloo():
+    outs = _run_spmd("encoder")
+    assert outs[0]["shape"] == outs[1]["shape"]
+    assert outs[0]["shape"][0] == 80
