@@ -341,47 +341,12 @@ def _sgd_step_torch(Xaug, y_float, fold_id, idx, W, V, spec, loss_id,
 # batched scoring
 # --------------------------------------------------------------------- #
 
-def batched_scores(ds, spec, W, loss_id, col_model, n_models, n_classes,
-                   metric, chunk=262144, force_eager=False):
-    """Per-model test-fold metric for every model in the batch.
-
-    ``col_model`` maps each column to its model id; a model owns 1 column
-    (binary / regression) or ``n_classes`` consecutive columns (internal
-    OvR).  The test rows of model m are those with
-    ``fold_id[i] == col_fold[first column of m]``.
-
-    Streams X in row chunks: Z-chunk GEMM → per-model sufficient
-    statistics (correct counts / confusion / log-loss sums / AUC
-    histograms), so the n×ncols score matrix never materializes.
-    Returns np.ndarray [n_models] of metric values.
-    """
-    device = ds.device
-    n = ds.n
-    ncols = spec.ncols
-    first_col = torch.as_tensor(
-        np.arange(n_models, dtype=np.int64) * (n_classes if n_classes > 2 else 1),
-        device=device,
-    )
-    model_fold = spec.col_fold[first_col]          # [n_models]
-
-    stats = _MetricState(metric, n_models, n_classes, device)
-    Wc = W.to(ds.comp_dtype)
-    for start in range(0, n, chunk):
-        end = min(n, start + chunk)
-        Xb = ds.Xaug[start:end]
-        Z = (Xb @ Wc).to(torch.float32)            # [m, ncols]
-        fid = ds.fold_id[start:end]
-        yb = ds.y_float[start:end]
-        stats.update(Z, yb, fid, spec, model_fold, n_classes)
-    return stats.finalize()
-
-
 def batched_scores_by_fold(ds, W, model_folds, col_class, n_classes,
                            metric, chunk=1 << 20):
     """Per-model test-fold metrics, fold-grouped (the fast search path).
 
-    Instead of streaming ALL rows against ALL columns with a test mask
-    (``batched_scores``), each fold's models are scored only on that
+    Instead of streaming all rows against all columns with a test
+    mask, each fold's models are scored only on that
     fold's rows: the scoring GEMM shrinks from n × total-cols to
     Σ_f |fold f| × (cols of fold f) — 1/n_folds of the work, and the
     mask logic disappears.  Models with fold -2 (full-data refit columns)

@@ -24,7 +24,6 @@ from ._sgd import (
     LOSS_SQUARED,
     ColumnSpec,
     DeviceDataset,
-    batched_scores,
     batched_scores_by_fold,
     batched_sgd_fit,
 )
