@@ -538,16 +538,18 @@ class ForestBuilder:
         return np.maximum(stats[:, 2] / w - mean * mean, 0.0)
 
     def _initial_sample_idx(self, weights):
-        """Pack rows with weight > 0 to the front of each tree's row."""
+        """Pack rows with weight > 0 to the front of each tree's row
+        (one vectorized nonzero over the whole batch)."""
         ds = self.ds
         TB = weights.shape[0]
         si = torch.zeros(TB, ds.n, dtype=torch.int32, device=ds.device)
-        counts = np.empty(TB, dtype=np.int64)
-        for t in range(TB):
-            nz = torch.nonzero(weights[t] > 0, as_tuple=False).flatten()
-            counts[t] = len(nz)
-            si[t, : len(nz)] = nz.to(torch.int32)
-        return si, counts
+        nz = torch.nonzero(weights > 0)          # [(t, i)] sorted
+        t_idx = nz[:, 0]
+        counts_t = torch.bincount(t_idx, minlength=TB)
+        offs = torch.cumsum(counts_t, 0) - counts_t
+        pos = torch.arange(len(nz), device=ds.device) - offs[t_idx]
+        si[t_idx, pos] = nz[:, 1].to(torch.int32)
+        return si, counts_t.cpu().numpy().astype(np.int64)
 
     def _impurity_np(self, stats, w):
         if w <= 0:
