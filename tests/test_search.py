@@ -137,3 +137,24 @@ def test_preds_attribute(small_xy):
     )
     gs.fit(X, y)
     assert gs.preds_.shape[0] == len(y)
+
+
+def test_non_partition_cv_falls_back(small_xy):
+    """ShuffleSplit folds don't partition the rows -> the batched device
+    solve must fall back to the generic per-task path (same results
+    contract)."""
+    from sklearn.model_selection import ShuffleSplit
+
+    from skdist_amd import Cluster
+    from skdist_amd.models import LogisticRegression
+
+    X, y = small_xy
+    gs = DistGridSearchCV(
+        LogisticRegression(epochs=10, random_state=0),
+        {"C": [0.1, 1.0]},
+        cv=ShuffleSplit(n_splits=3, test_size=0.3, random_state=0),
+        sc=Cluster(),
+    )
+    gs.fit(X, y)
+    assert gs.best_score_ > 0.7
+    assert len(gs.cv_results_["mean_test_score"]) == 2
