@@ -475,9 +475,42 @@ class DistMultiModelSearch(BaseEstimator, MetaEstimatorMixin):
             ):
                 entries.append((mi, pi, params))
 
+        # per-model batched fast path: a model family exposing the
+        # batched protocol solves ALL its sampled param sets as columns
+        # of one device solve; only the rest go through the task pool
+        agg = {}
+        batched_mis = set()
+        if sc is not None and not fit_params:
+            from ..models.linear import FallbackToGeneric
+
+            for mi, (name, est, dists, n_override) in enumerate(models):
+                if not hasattr(est, "batched_cv_fit_score"):
+                    continue
+                cand = [p for (m2, pi, p) in entries if m2 == mi]
+                if not cand:
+                    continue
+                try:
+                    out = est.batched_cv_fit_score(
+                        np.asarray(X), None if y is None else np.asarray(y),
+                        candidate_params=cand, cv_splits=folds,
+                        scoring=self.scoring, scorers=None, cluster=sc,
+                    )
+                except FallbackToGeneric:
+                    continue
+                tasks_out = out["tasks"]
+                nf = len(folds)
+                for ci in range(len(cand)):
+                    agg[(mi, ci)] = [
+                        tasks_out[ci * nf + fi]["test_scores"]["score"]
+                        for fi in range(nf)
+                    ]
+                batched_mis.add(mi)
+
         tasks = []
         tid = 0
         for mi, pi, params in entries:
+            if mi in batched_mis:
+                continue
             for fi, split in enumerate(folds):
                 tasks.append((tid, mi, pi, params, split))
                 tid += 1
@@ -497,7 +530,9 @@ class DistMultiModelSearch(BaseEstimator, MetaEstimatorMixin):
             scorer = check_scoring(est, scoring=self.scoring)
             return tid_, mi, pi, float(scorer(est, X_te, y_te))
 
-        if sc is None:
+        if not tasks:
+            results = []
+        elif sc is None:
             results = run_local_tasks(
                 task_fn, tasks, n_jobs=self.n_jobs,
                 pre_dispatch=self.pre_dispatch,
@@ -506,7 +541,6 @@ class DistMultiModelSearch(BaseEstimator, MetaEstimatorMixin):
             results = sc.run_tasks(task_fn, tasks)
 
         # aggregate mean score per (model, params)
-        agg = {}
         for tid_, mi, pi, score in results:
             agg.setdefault((mi, pi), []).append(score)
         rows = []

@@ -54,3 +54,36 @@ def test_multimodel_continuous_dists():
     )
     mm.fit(X, y)
     assert len(mm.cv_results_["params"]) == 3
+
+
+def test_multimodel_batched_path_matches_generic():
+    """With a Cluster, our linear families solve all their sampled
+    param sets as one batched solve per model; results must agree with
+    the generic per-task path."""
+    from skdist_amd import Cluster
+    from skdist_amd.models import LinearSVC, LogisticRegression
+
+    rng = np.random.default_rng(0)
+    X = rng.standard_normal((1200, 10)).astype(np.float32)
+    y = ((X[:, :4] @ rng.standard_normal(4)) > 0).astype(np.int64)
+    models = [
+        ("lr", LogisticRegression(epochs=10, random_state=0),
+         {"C": [0.1, 1.0, 10.0]}),
+        ("svc", LinearSVC(epochs=10, random_state=0),
+         {"C": [0.1, 1.0]}),
+    ]
+    mm_b = DistMultiModelSearch(models, n=2, cv=3, sc=Cluster(),
+                                random_state=0)
+    mm_b.fit(X, y)
+    mm_g = DistMultiModelSearch(models, n=2, cv=3, sc=None,
+                                random_state=0)
+    mm_g.fit(X, y)
+    assert mm_b.best_model_name_ in ("lr", "svc")
+    assert len(mm_b.cv_results_["params"]) == len(
+        mm_g.cv_results_["params"])
+    # same sampled candidates, closely matching scores
+    assert mm_b.cv_results_["params"] == mm_g.cv_results_["params"]
+    np.testing.assert_allclose(
+        mm_b.cv_results_["mean_test_score"],
+        mm_g.cv_results_["mean_test_score"], atol=0.05)
+    assert mm_b.best_score_ > 0.8
