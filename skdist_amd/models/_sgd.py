@@ -365,6 +365,12 @@ def batched_scores_by_fold(ds, W, model_folds, col_class, n_classes,
     class _Shim:
         pass
 
+    hip_modes = {"accuracy": 0, "r2": 1, "neg_mean_squared_error": 1}
+    use_hip_kernel = (
+        _use_hip(device)
+        and metric in hip_modes
+        and (not ds.is_cls or n_classes == 2)
+    )
     for f in np.unique(model_folds[model_folds >= 0]):
         mids = np.flatnonzero(model_folds == f)
         cols_np = (mids[:, None] * cpm + np.arange(cpm)).ravel()
@@ -372,6 +378,10 @@ def batched_scores_by_fold(ds, W, model_folds, col_class, n_classes,
         Wf = Wc.index_select(1, cols).contiguous()
         rows = torch.nonzero(ds.fold_id == int(f)).flatten()
         nm = len(mids)
+        if use_hip_kernel:
+            out[mids] = _score_fold_hip(
+                ds, Wf, rows, hip_modes[metric], metric)
+            continue
         spec = _Shim()
         spec.col_class = torch.as_tensor(
             np.ascontiguousarray(col_class[cols_np]), device=device)
@@ -387,6 +397,38 @@ def batched_scores_by_fold(ds, W, model_folds, col_class, n_classes,
             state.update(Z, yb, fid, spec, model_fold_t, n_classes)
         out[mids] = state.finalize()
     return out
+
+
+def _score_fold_hip(ds, Wf, rows, mode, metric):
+    """One fold through the fused scoring kernel (k_score,
+    sgd_kernels.hip): gathered 128-padded test rows, per-column
+    sufficient statistics accumulated in the GEMM epilogue."""
+    from ..ops import require_hip
+
+    device = ds.device
+    m = len(rows)
+    fa = ds.Xaug.shape[1]
+    ncols = Wf.shape[1]
+    ncp = (ncols + 127) // 128 * 128
+    m_pad = max(128, (m + 127) // 128 * 128)
+    Xb = torch.zeros(m_pad, fa, dtype=ds.comp_dtype, device=device)
+    torch.index_select(ds.Xaug, 0, rows, out=Xb[:m])
+    yf = torch.full((m_pad,), -1e30, dtype=torch.float32, device=device)
+    yf[:m] = ds.y_float.index_select(0, rows)
+    WbfT = torch.zeros(ncp, fa, dtype=ds.comp_dtype, device=device)
+    WbfT[:ncols] = Wf.t()
+    nstat = 2 if mode == 0 else 3
+    stats = torch.zeros(ncp * nstat, dtype=torch.float32, device=device)
+    require_hip().score_fold(Xb.contiguous(), WbfT.contiguous(), yf,
+                             stats, mode)
+    sh = stats.cpu().numpy().reshape(ncp, nstat)[:ncols]
+    if mode == 0:
+        return sh[:, 0] / np.clip(sh[:, 1], 1, None)
+    sse, sy, syy = sh[:, 0], sh[:, 1], sh[:, 2]
+    if metric == "neg_mean_squared_error":
+        return -sse / max(m, 1)
+    sst = syy - sy * sy / max(m, 1)
+    return 1.0 - sse / np.clip(sst, 1e-12, None)
 
 
 class _MetricState:

@@ -144,6 +144,10 @@ extern "C" hipError_t skdist_forest_apply(
     const void* X, const void* feat, const void* thr, const void* left,
     const void* right, const void* roots, void* out_leaf, long long rows,
     int f, int n_trees, hipStream_t stream);
+extern "C" hipError_t skdist_score(
+    const void* Xf, const void* WbfT, const void* yf, void* out,
+    long long m_pad, int fa, int ncols_pad, int mode,
+    hipStream_t stream);
 extern "C" hipError_t skdist_standardize(
     const void* X, const void* mean, const void* inv_std, void* out,
     long long n, int f, int fa, hipStream_t stream);
@@ -255,6 +259,23 @@ void forest_apply(torch::Tensor X, torch::Tensor feat, torch::Tensor thr,
                 hipGetErrorString(err));
 }
 
+void score_fold(torch::Tensor Xf, torch::Tensor WbfT, torch::Tensor yf,
+                torch::Tensor out, int64_t mode) {
+    for (auto* t : {&Xf, &WbfT, &yf, &out}) {
+        CHECK_DEV(*t);
+        CHECK_CONT(*t);
+    }
+    TORCH_CHECK(Xf.scalar_type() == torch::kBFloat16, "Xf must be bf16");
+    TORCH_CHECK(Xf.size(0) % 128 == 0, "m_pad must be a mult of 128");
+    TORCH_CHECK(WbfT.size(0) % 128 == 0, "ncols_pad mult of 128");
+    auto stream = c10::hip::getCurrentHIPStream().stream();
+    hipError_t err = skdist_score(
+        Xf.data_ptr(), WbfT.data_ptr(), yf.data_ptr(), out.data_ptr(),
+        Xf.size(0), (int)Xf.size(1), (int)WbfT.size(0), (int)mode,
+        stream);
+    TORCH_CHECK(err == hipSuccess, "score_fold: ", hipGetErrorString(err));
+}
+
 void standardize(torch::Tensor X, torch::Tensor mean,
                  torch::Tensor inv_std, torch::Tensor out, int64_t f) {
     for (auto* t : {&X, &mean, &inv_std, &out}) {
@@ -295,6 +316,8 @@ void hash_vectorize(torch::Tensor bytes, torch::Tensor doc_off,
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("sgd_step", &sgd_step, "fused batched SGD step (K1+K2+K3)");
     m.def("sgd_epoch", &sgd_epoch, "one epoch of fused SGD steps");
+    m.def("score_fold", &score_fold,
+          "fused fold-scoring GEMM + per-column stats");
     m.def("standardize", &standardize,
           "fused standardize + augment + bf16 cast");
     m.def("hash_vectorize", &hash_vectorize,

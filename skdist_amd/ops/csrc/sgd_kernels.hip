@@ -404,3 +404,151 @@ extern "C" hipError_t skdist_standardize(
                        (const float*)inv_std, (__bf16*)out, n, f, fa);
     return hipGetLastError();
 }
+
+// ---------------------------------------------------------------------- //
+// K5: fused test-fold scoring — GEMM tile + per-column sufficient stats
+// (the k-fold CV scoring kernel; SURVEY.md §2.4 "k-fold split + score").
+// The caller gathers ONE fold's test rows into a 128-padded bf16 buffer
+// (pad rows zero, y pad = -1e30 sentinel -> excluded), so no masks are
+// needed: every row scores every column.
+//   mode 0 (binary accuracy): out[col*2]   += #(sign(z) == y)
+//                             out[col*2+1] += #rows           (valid)
+//   mode 1 (regression):      out[col*3]   += (z-y)^2
+//                             out[col*3+1] += y, out[col*3+2] += y^2
+// grid: (m_pad/128, ncols_pad/128), block 256 (4 waves of 64x64).
+// ---------------------------------------------------------------------- //
+extern "C" __global__ __launch_bounds__(256) void k_score(
+    const __bf16* __restrict__ Xf,    // [m_pad][fa] gathered fold rows
+    const __bf16* __restrict__ WbfT,  // [ncols_pad][fa]
+    const float* __restrict__ yf,     // [m_pad] (-1e30 pad sentinel)
+    float* __restrict__ out,          // [ncols_pad * (2|3)]
+    int fa, int ncols_pad, int mode)
+{
+    extern __shared__ __attribute__((aligned(16))) char sm[];
+    #define sbufA(b) ((__bf16*)(sm + (b) * 16384))
+    #define sbufB(b) ((__bf16*)(sm + 8192 + (b) * 16384))
+    float* y_s = (float*)(sm + 32768);        // [128]
+    float* red = (float*)(sm + 33280);        // [3][132] per-col partials
+
+    const int tid = threadIdx.x;
+    const int bm = blockIdx.x * 128;
+    const int bn = blockIdx.y * 128;
+    const int lane = tid & 63;
+    const int w = tid >> 6;
+    if (tid < 128) y_s[tid] = yf[bm + tid];
+
+    const __bf16* Abase = Xf + (long long)bm * fa;
+    const __bf16* Bbase = WbfT + (long long)bn * fa;
+    const int wr = ((w >> 1) & 1) * 64;
+    const int wc = (w & 1) * 64;
+    const int fr = lane & 15;
+    const int fk = (lane >> 4) * 8;
+
+    f32x4 acc[4][4] = {};
+    const int nk = fa / BK;
+    int cur = 0;
+    {   // stage first tiles (STAGE_TILE expects `row`, `lane`, `w`)
+        _Pragma("unroll") for (int i = 0; i < 2; ++i) {
+            const int c = w + 4 * i;
+            const int row = 16 * c + (lane >> 2);
+            const int slot = (lane & 3) ^ ((row >> 2) & 3);
+            __builtin_amdgcn_global_load_lds(
+                (const unsigned int*)(Abase + (long long)row * fa +
+                                      slot * 8),
+                (unsigned int*)(sbufA(0) + c * 512), 16, 0, 0);
+            __builtin_amdgcn_global_load_lds(
+                (const unsigned int*)(Bbase + (long long)row * fa +
+                                      slot * 8),
+                (unsigned int*)(sbufB(0) + c * 512), 16, 0, 0);
+        }
+    }
+    for (int kt = 0; kt < nk; ++kt) {
+        __syncthreads();
+        if (kt + 1 < nk) {
+            _Pragma("unroll") for (int i = 0; i < 2; ++i) {
+                const int c = w + 4 * i;
+                const int row = 16 * c + (lane >> 2);
+                const int slot = (lane & 3) ^ ((row >> 2) & 3);
+                __builtin_amdgcn_global_load_lds(
+                    (const unsigned int*)(Abase + (long long)row * fa +
+                                          (kt + 1) * BK + slot * 8),
+                    (unsigned int*)(sbufA(cur ^ 1) + c * 512), 16, 0, 0);
+                __builtin_amdgcn_global_load_lds(
+                    (const unsigned int*)(Bbase + (long long)row * fa +
+                                          (kt + 1) * BK + slot * 8),
+                    (unsigned int*)(sbufB(cur ^ 1) + c * 512), 16, 0, 0);
+            }
+        }
+        bf16x8 af[4], bfr[4];
+        #pragma unroll
+        for (int mi = 0; mi < 4; ++mi)
+            af[mi] = frag_load(sbufA(cur), wr + mi * 16 + fr, fk);
+        #pragma unroll
+        for (int ni = 0; ni < 4; ++ni)
+            bfr[ni] = frag_load(sbufB(cur), wc + ni * 16 + fr, fk);
+        #pragma unroll
+        for (int mi = 0; mi < 4; ++mi)
+            #pragma unroll
+            for (int ni = 0; ni < 4; ++ni)
+                acc[mi][ni] =
+                    MFMA_BF16_16x16x32(af[mi], bfr[ni], acc[mi][ni]);
+        cur ^= 1;
+    }
+    __syncthreads();
+
+    // per-lane partials over this wave's 16 rows x 4 col-groups
+    const int rw = lane >> 4;
+    float s0[4] = {}, s1[4] = {}, s2[4] = {};
+    #pragma unroll
+    for (int mi = 0; mi < 4; ++mi) {
+        #pragma unroll
+        for (int ni = 0; ni < 4; ++ni) {
+            #pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                const int row = wr + mi * 16 + rw * 4 + r;
+                const float yv = y_s[row];
+                if (yv < -1e29f) continue;   // pad row
+                const float z = acc[mi][ni][r];
+                if (mode == 0) {
+                    s0[ni] += ((z >= 0.f) == (yv != 0.f)) ? 1.f : 0.f;
+                    s1[ni] += 1.f;
+                } else {
+                    const float e = z - yv;
+                    s0[ni] += e * e;
+                    s1[ni] += yv;
+                    s2[ni] += yv * yv;
+                }
+            }
+        }
+    }
+    // reduce across the 16 lanes-groups sharing a column (rows axis):
+    // lanes with same (lane&15, wc) hold different rows; use LDS.
+    const int nstat = (mode == 0) ? 2 : 3;
+    for (int st = 0; st < nstat; ++st) {
+        __syncthreads();
+        for (int i = tid; i < 132; i += 256) red[i] = 0.f;
+        __syncthreads();
+        #pragma unroll
+        for (int ni = 0; ni < 4; ++ni) {
+            const float v = (st == 0) ? s0[ni] : (st == 1 ? s1[ni]
+                                                          : s2[ni]);
+            atomicAdd(&red[wc + ni * 16 + fr], v);
+        }
+        __syncthreads();
+        for (int c = tid; c < 128; c += 256)
+            atomicAdd(&out[(long long)(bn + c) * nstat + st], red[c]);
+    }
+}
+
+extern "C" hipError_t skdist_score(
+    const void* Xf, const void* WbfT, const void* yf, void* out,
+    long long m_pad, int fa, int ncols_pad, int mode,
+    hipStream_t stream)
+{
+    dim3 grid((unsigned)(m_pad / 128), ncols_pad / 128);
+    hipLaunchKernelGGL(k_score, grid, dim3(256), 33280 + 132 * 4, stream,
+                       (const __bf16*)Xf, (const __bf16*)WbfT,
+                       (const float*)yf, (float*)out, fa, ncols_pad,
+                       mode);
+    return hipGetLastError();
+}
