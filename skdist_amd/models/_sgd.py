@@ -369,7 +369,7 @@ def batched_scores_by_fold(ds, W, model_folds, col_class, n_classes,
     use_hip_kernel = (
         _use_hip(device)
         and metric in hip_modes
-        and (not ds.is_cls or n_classes == 2)
+        and (ds.classes_ is None or n_classes == 2)
     )
     for f in np.unique(model_folds[model_folds >= 0]):
         mids = np.flatnonzero(model_folds == f)
