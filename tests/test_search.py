@@ -158,3 +158,19 @@ def test_non_partition_cv_falls_back(small_xy):
     gs.fit(X, y)
     assert gs.best_score_ > 0.7
     assert len(gs.cv_results_["mean_test_score"]) == 2
+
+
+def test_out_of_fold_preds(small_xy):
+    """preds=True returns out-of-fold predictions row-aligned with X
+    (reference search.py:551-560)."""
+    from skdist_amd.models import LogisticRegression
+
+    X, y = small_xy
+    gs = DistGridSearchCV(
+        LogisticRegression(epochs=10, random_state=0),
+        {"C": [0.1, 1.0]}, cv=3, preds=True, sc=None)
+    gs.fit(X, y)
+    assert gs.preds_.shape[0] == len(y)
+    # proba columns are row-aligned: argmax should mostly match labels
+    agree = (gs.classes_[gs.preds_.argmax(axis=1)] == y).mean()
+    assert agree > 0.8
