@@ -71,7 +71,7 @@ class DeviceDataset:
     """
 
     def __init__(self, X, y, cluster=None, device=None, standardize=True,
-                 classes=None):
+                 classes=None, sample_weight=None):
         self.cluster = cluster
         if device is None:
             device = cluster.device if cluster is not None else (
@@ -160,6 +160,13 @@ class DeviceDataset:
             self.Xaug = torch.cat(cols, dim=1).to(comp_dtype).contiguous()
         del Xt
 
+        if sample_weight is not None:
+            self.row_w = torch.as_tensor(
+                np.ascontiguousarray(sample_weight, dtype=np.float32),
+                device=self.device)
+        else:
+            self.row_w = None
+
         self.fold_id = None  # set by set_cv_partition
 
     def set_cv_partition(self, cv_splits):
@@ -204,7 +211,11 @@ class DeviceDataset:
         folds = (
             self.fold_id.index_select(0, perm).to(torch.int32).contiguous()
         )
-        self._shuf = (Xs, XsT, ys, folds)
+        rw = (
+            self.row_w.index_select(0, perm).contiguous()
+            if self.row_w is not None else None
+        )
+        self._shuf = (Xs, XsT, ys, folds, rw)
         self._shuf_key = key
         return self._shuf
 
@@ -280,6 +291,7 @@ def batched_sgd_fit(ds, spec, loss, epochs, batch_size, seed=0,
             _sgd_step_torch(
                 ds.Xaug, ds.y_float, ds.fold_id, idx, W, V, spec,
                 loss_id, lr_scale, momentum, ds.intercept_row,
+                row_w=ds.row_w,
             )
             if fmask is not None:
                 W.mul_(fmask)
@@ -289,7 +301,7 @@ def batched_sgd_fit(ds, spec, loss, epochs, batch_size, seed=0,
 
 
 def _sgd_step_torch(Xaug, y_float, fold_id, idx, W, V, spec, loss_id,
-                    lr_scale, momentum, intercept_row=None):
+                    lr_scale, momentum, intercept_row=None, row_w=None):
     """One mini-batch step, eager torch — the numerics reference the HIP
     kernels are tested against (fp32 on CPU; on GPU it mirrors the kernel's
     bf16-in/fp32-accumulate)."""
@@ -322,8 +334,14 @@ def _sgd_step_torch(Xaug, y_float, fold_id, idx, W, V, spec, loss_id,
             | (yb.unsqueeze(1) == c2.float())
         )
         G = G * (mask & pair_ok).to(torch.float32)
+    denom = float(m)
+    if row_w is not None:
+        w = row_w[idx]
+        G = G * w.unsqueeze(1)
+        denom = max(float(w.sum()), 1e-30)
 
-    grad = (Xb.transpose(0, 1).to(comp) @ G.to(comp)).to(torch.float32) / m
+    grad = (Xb.transpose(0, 1).to(comp) @ G.to(comp)).to(
+        torch.float32) / denom
     # L2 on weights only, not the intercept row
     ir = W.shape[0] - 1 if intercept_row is None else intercept_row
     l2 = spec.col_l2.unsqueeze(0) * W

@@ -57,8 +57,6 @@ class _BatchedLinearBase(BaseEstimator):
 
     # ------------------------------------------------------------------ #
     def fit(self, X, y, sample_weight=None):
-        if sample_weight is not None:
-            raise ValueError("sample_weight is not supported yet")
         t0 = time.perf_counter()
         sc = getattr(self, "sc", None)
         is_clf = isinstance(self, ClassifierMixin)
@@ -67,6 +65,7 @@ class _BatchedLinearBase(BaseEstimator):
             cluster=None,  # single fit: local device, no broadcast
             device=sc.device if sc is not None else None,
             standardize=self.standardize,
+            sample_weight=sample_weight,
         )
         ds.set_cv_partition([])  # no CV mask
         if is_clf:

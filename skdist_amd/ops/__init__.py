@@ -177,12 +177,24 @@ def hip_sgd_solve(ds, spec, loss_id, epochs, batch_size, seed, momentum,
     perm = torch.as_tensor(
         rng.permutation(n), dtype=torch.int64, device=device
     )
-    Xs, XsT, ys, folds = ds.shuffled_views(perm)
+    Xs, XsT, ys, folds, rw = ds.shuffled_views(perm)
+    if rw is None:
+        rw_t = torch.empty(0, dtype=torch.float32, device=device)
+        inv_m = torch.tensor(
+            [1.0 / min(bs, n - s) for s in range(0, n, bs)],
+            dtype=torch.float32)
+    else:
+        rw_t = rw
+        sums = [
+            float(rw[s: s + bs].sum()) for s in range(0, n, bs)
+        ]
+        inv_m = torch.tensor(
+            [1.0 / max(v, 1e-30) for v in sums], dtype=torch.float32)
     for epoch in range(epochs):
         lr_scale = 1.0 / (1.0 + lr_decay * epoch)
         ext.sgd_epoch(
             Xs, XsT, GT, W, V, WbfT, partial, ys, folds,
-            cls_p, cfold_p, cls2_p, lr_p, l2_p, fmask,
+            cls_p, cfold_p, cls2_p, lr_p, l2_p, fmask, rw_t, inv_m,
             bs, int(loss_id), float(lr_scale), float(momentum),
             int(ds.intercept_row),
         )

@@ -11,6 +11,7 @@ extern "C" hipError_t skdist_sgd_step(
     const void* y, const void* fold,
     const void* col_class, const void* col_fold, const void* col_class2,
     const void* col_lr, const void* col_l2, const void* fmask,
+    const void* row_w, float inv_m,
     long long start, long long m, long long n, long long n_pad,
     long long fa_store, int fa, int ncols_pad,
     int gt_stride, int splitk, int loss_id,
@@ -53,9 +54,9 @@ void sgd_step(torch::Tensor Xs, torch::Tensor XsT, torch::Tensor GT,
               torch::Tensor col_class, torch::Tensor col_fold,
               torch::Tensor col_class2,
               torch::Tensor col_lr, torch::Tensor col_l2,
-              torch::Tensor fmask,
+              torch::Tensor fmask, torch::Tensor row_w,
               int64_t start, int64_t m, int64_t loss_id, double lr_scale,
-              double momentum, int64_t intercept_row) {
+              double momentum, int64_t intercept_row, double inv_m) {
     for (auto* t : {&Xs, &XsT, &GT, &W, &WbfT, &partial, &y, &fold,
                     &col_class, &col_fold, &col_lr, &col_l2}) {
         CHECK_DEV(*t);
@@ -71,6 +72,7 @@ void sgd_step(torch::Tensor Xs, torch::Tensor XsT, torch::Tensor GT,
         col_class.data_ptr(), col_fold.data_ptr(), col_class2.data_ptr(),
         col_lr.data_ptr(), col_l2.data_ptr(),
         fmask.numel() ? fmask.data_ptr() : nullptr,
+        row_w.numel() ? row_w.data_ptr() : nullptr, (float)inv_m,
         start, m, a.n, a.n_pad, a.fa_store, (int)a.fa,
         (int)a.ncols_pad, (int)a.gt_stride, (int)a.splitk, (int)loss_id,
         (float)lr_scale, (float)momentum, (int)intercept_row, stream);
@@ -86,14 +88,18 @@ void sgd_epoch(torch::Tensor Xs, torch::Tensor XsT, torch::Tensor GT,
                torch::Tensor col_class, torch::Tensor col_fold,
                torch::Tensor col_class2,
                torch::Tensor col_lr, torch::Tensor col_l2,
-               torch::Tensor fmask,
+               torch::Tensor fmask, torch::Tensor row_w,
+               torch::Tensor inv_m_cpu,  // [n_batches] f32 on CPU
                int64_t batch_size, int64_t loss_id, double lr_scale,
                double momentum, int64_t intercept_row) {
     auto a = check_args(Xs, XsT, GT, W, partial, y);
     TORCH_CHECK(batch_size % 128 == 0, "batch_size must be a mult of 128");
     const bool has_V = V.numel() > 0;
+    TORCH_CHECK(!inv_m_cpu.is_cuda(), "inv_m must be a CPU tensor");
+    const float* inv_p = (const float*)inv_m_cpu.data_ptr();
     auto stream = c10::hip::getCurrentHIPStream().stream();
-    for (int64_t start = 0; start < a.n; start += batch_size) {
+    int64_t bi = 0;
+    for (int64_t start = 0; start < a.n; start += batch_size, ++bi) {
         const int64_t m = std::min(batch_size, a.n - start);
         hipError_t err = skdist_sgd_step(
             Xs.data_ptr(), XsT.data_ptr(), WbfT.data_ptr(), GT.data_ptr(),
@@ -102,6 +108,7 @@ void sgd_epoch(torch::Tensor Xs, torch::Tensor XsT, torch::Tensor GT,
             col_class.data_ptr(), col_fold.data_ptr(),
             col_class2.data_ptr(), col_lr.data_ptr(), col_l2.data_ptr(),
             fmask.numel() ? fmask.data_ptr() : nullptr,
+            row_w.numel() ? row_w.data_ptr() : nullptr, inv_p[bi],
             start, m, a.n, a.n_pad, a.fa_store,
             (int)a.fa, (int)a.ncols_pad, (int)a.gt_stride, (int)a.splitk,
             (int)loss_id, (float)lr_scale, (float)momentum,

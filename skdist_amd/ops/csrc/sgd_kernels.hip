@@ -67,6 +67,7 @@ extern "C" __global__ __launch_bounds__(256) void k_fwd_gt(
     const int* __restrict__ col_class,
     const int* __restrict__ col_fold,
     const int* __restrict__ col_class2,  // >=0: one-vs-one partner class
+    const float* __restrict__ row_w,     // per-row sample weights or null
     int start, int m, long long n, int fa, int ncols_pad, int gt_stride,
     int loss_id)
 {
@@ -80,6 +81,7 @@ extern "C" __global__ __launch_bounds__(256) void k_fwd_gt(
     int* cls_s = (int*)(meta + 1024);       // [BN]
     int* cfold_s = (int*)(meta + 1536);     // [BN]
     int* cls2_s = (int*)(meta + 2048);      // [BN]
+    float* rw_s = (float*)(meta + 2560);    // [BM]
 
     const int tid = threadIdx.x;
     const int bm = blockIdx.x * BM;
@@ -92,6 +94,7 @@ extern "C" __global__ __launch_bounds__(256) void k_fwd_gt(
         const bool ok = r < n;
         y_s[tid] = ok ? y[r] : 0.f;
         fold_s[tid] = ok ? fold[r] : -9;
+        rw_s[tid] = (row_w != nullptr && ok) ? row_w[r] : 1.f;
     } else {
         const int c = tid - BM;
         cls_s[c] = col_class[bn + c];
@@ -156,7 +159,7 @@ extern "C" __global__ __launch_bounds__(256) void k_fwd_gt(
                 const float yv = y_s[row];
                 const float t =
                     (cls < 0) ? yv : (yv == (float)cls ? 1.f : 0.f);
-                const float g = dloss(loss_id, z, t);
+                const float g = dloss(loss_id, z, t) * rw_s[row];
                 bool train = (fold_s[row] != cfo) && (bm + row < m);
                 if (c2 >= 0)  // one-vs-one: only the pair's rows train
                     train = train &&
@@ -310,6 +313,7 @@ extern "C" hipError_t skdist_sgd_step(
     const void* y, const void* fold,
     const void* col_class, const void* col_fold, const void* col_class2,
     const void* col_lr, const void* col_l2, const void* fmask,
+    const void* row_w, float inv_m,
     long long start, long long m, long long n, long long n_pad,
     long long fa_store, int fa, int ncols_pad,
     int gt_stride, int splitk, int loss_id,
@@ -319,12 +323,12 @@ extern "C" hipError_t skdist_sgd_step(
     const int m_pad = (int)((m + BM - 1) / BM) * BM;
     {
         dim3 grid(m_pad / BM, ncols_pad / BN);
-        size_t lds = (size_t)BN * LDC * 2 + 2560;
+        size_t lds = (size_t)BN * LDC * 2 + 3072;
         hipLaunchKernelGGL(k_fwd_gt, grid, dim3(256), lds, stream,
                            (const __bf16*)Xs, (const __bf16*)WbfT_in,
                            (__bf16*)GT, (const float*)y, (const int*)fold,
                            (const int*)col_class, (const int*)col_fold,
-                           (const int*)col_class2,
+                           (const int*)col_class2, (const float*)row_w,
                            (int)start, (int)m, n, fa, ncols_pad, gt_stride,
                            loss_id);
         HIP_CHECK(hipGetLastError());
@@ -345,8 +349,7 @@ extern "C" hipError_t skdist_sgd_step(
                            (const float*)partial, (float*)W, (float*)V,
                            (__bf16*)WbfT, (const float*)col_lr,
                            (const float*)col_l2,
-                           (const unsigned char*)fmask,
-                           (float)(1.0 / (double)m),
+                           (const unsigned char*)fmask, inv_m,
                            lr_scale, momentum, intercept_row, splitk, fa,
                            ncols_pad);
         HIP_CHECK(hipGetLastError());

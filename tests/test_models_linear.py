@@ -114,3 +114,37 @@ def test_batched_accuracy_default_scoring():
     gs = DistGridSearchCV(est, {"C": [0.1, 1.0]}, cv=3, sc=Cluster())
     gs.fit(X, y)
     assert gs.best_score_ > 0.9
+
+
+def test_sample_weight_zero_rows_have_no_influence():
+    """Rows with sample_weight 0 must not affect the solve at all: their
+    labels can be arbitrary (exact equality; weights flow through the
+    fused gradient epilogue)."""
+    rng = np.random.default_rng(0)
+    X = rng.standard_normal((1000, 8)).astype(np.float32)
+    y = ((X[:, :3] @ rng.standard_normal(3)) > 0).astype(np.int64)
+    w = np.ones(1000, dtype=np.float64)
+    w[::5] = 0.0
+    y_bad = y.copy()
+    y_bad[::5] = 1 - y_bad[::5]  # garbage labels on zero-weight rows
+
+    a = LogisticRegression(epochs=8, random_state=0).fit(
+        X, y, sample_weight=w)
+    b = LogisticRegression(epochs=8, random_state=0).fit(
+        X, y_bad, sample_weight=w)
+    np.testing.assert_allclose(a.coef_, b.coef_, atol=1e-7)
+    np.testing.assert_allclose(a.intercept_, b.intercept_, atol=1e-7)
+
+
+def test_sample_weight_upweights_rows():
+    rng = np.random.default_rng(1)
+    X = rng.standard_normal((600, 4)).astype(np.float32)
+    y = (X[:, 0] > 0).astype(np.int64)
+    # heavily upweight the rows where feature 1 decides the label
+    y2 = (X[:, 1] > 0).astype(np.int64)
+    w = np.where(y == y2, 10.0, 0.1)
+    m = LogisticRegression(epochs=15, random_state=0).fit(
+        X, y, sample_weight=w)
+    m0 = LogisticRegression(epochs=15, random_state=0).fit(X, y)
+    # weighting toward agreement rows grows |coef| on feature 1
+    assert abs(m.coef_[0][1]) > abs(m0.coef_[0][1])
