@@ -160,7 +160,7 @@ class DistBaseSearchCV(BaseEstimator, MetaEstimatorMixin, metaclass=ABCMeta):
             sc is not None
             and hasattr(base_estimator, "batched_cv_fit_score")
             and not self.preds
-            and not fit_params
+            and set(fit_params) <= {"sample_weight"}
         )
         out = None
         refit_fn = None
@@ -176,6 +176,7 @@ class DistBaseSearchCV(BaseEstimator, MetaEstimatorMixin, metaclass=ABCMeta):
                     scorers=scorers,
                     cluster=sc,
                     return_train_score=self.return_train_score,
+                    sample_weight=fit_params.get("sample_weight"),
                 )
             except FallbackToGeneric:
                 out = None

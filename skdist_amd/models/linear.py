@@ -131,7 +131,8 @@ class _BatchedLinearBase(BaseEstimator):
     # ------------------------------------------------------------------ #
     def batched_cv_fit_score(self, X, y, candidate_params, cv_splits,
                              scoring, scorers, cluster,
-                             return_train_score=False):
+                             return_train_score=False,
+                             sample_weight=None):
         """Train all (candidate × fold) models in one batched device solve.
 
         Returns the same per-task result dicts the generic path produces
@@ -154,6 +155,7 @@ class _BatchedLinearBase(BaseEstimator):
             None if X is None else np.asarray(X, dtype=np.float32),
             None if y is None else np.asarray(y),
             cluster=cluster, standardize=self.standardize,
+            sample_weight=sample_weight,
         )
         if not ds.set_cv_partition(cv_splits):
             raise FallbackToGeneric("cv splits do not partition the data")
