@@ -257,7 +257,34 @@ class DistBaseForest(BaseEstimator):
             )
 
     # ------------------------------------------------------------------ #
+    def _device_forest(self):
+        """Opportunistic batched-traversal scorer (k_forest_predict) for
+        HIP-fitted trees when a GPU is visible; never pickled."""
+        import torch
+
+        if not torch.cuda.is_available():
+            return None
+        if getattr(self, "_flat_cache", None) is None:
+            from ..models.forest import flat_forest_for
+
+            try:
+                self._flat_cache = flat_forest_for(self, "cuda") or False
+            except Exception:
+                self._flat_cache = False
+        return self._flat_cache or None
+
+    def __getstate__(self):
+        try:
+            state = dict(super().__getstate__())
+        except AttributeError:  # plain object protocol
+            state = dict(self.__dict__)
+        state.pop("_flat_cache", None)  # device tensors never pickle
+        return state
+
     def apply(self, X):
+        flat = self._device_forest()
+        if flat is not None:
+            return flat.apply(X)
         return np.column_stack([t.apply(X) for t in self.estimators_])
 
     @property
@@ -273,6 +300,9 @@ class _ForestClassifierMixin(ClassifierMixin):
     _is_classifier = True
 
     def predict_proba(self, X):
+        flat = self._device_forest()
+        if flat is not None:
+            return flat.predict_proba(X)
         proba = None
         for tree in self.estimators_:
             p = tree.predict_proba(X)
@@ -292,6 +322,9 @@ class _ForestRegressorMixin(RegressorMixin):
     _is_classifier = False
 
     def predict(self, X):
+        flat = self._device_forest()
+        if flat is not None:
+            return flat.predict_value(X)[:, 0]
         out = None
         for tree in self.estimators_:
             p = tree.predict(X)
