@@ -206,7 +206,21 @@ class DistOneVsRestClassifier(ClassifierMixin, BaseEstimator):
                 cols.append(est.decision_function(X))
         return np.column_stack(cols)
 
+    def _try_device(self, method, X):
+        """Opportunistic GPU scoring of the fitted model (one GEMM +
+        fused link/normalize) when a device is visible; None -> host."""
+        import torch
+
+        if not torch.cuda.is_available():
+            return None
+        fn = self._device_predict_fn(method, "cuda")
+        return None if fn is None else fn(X)
+
     def predict(self, X):
+        if self.mlb_ is None and not getattr(self, "multilabel_", False):
+            out = self._try_device("predict", X)
+            if out is not None:
+                return out
         scores = self._scores(X)
         if self.mlb_ is not None or getattr(self, "multilabel_", False):
             ind = (scores > 0.5).astype(int)
@@ -216,7 +230,12 @@ class DistOneVsRestClassifier(ClassifierMixin, BaseEstimator):
         return self.classes_[scores.argmax(axis=1)]
 
     def predict_proba(self, X):
-        """Vectorized replacement for reference multiclass.py:337-362."""
+        """Vectorized replacement for reference multiclass.py:337-362;
+        on a GPU machine the k binary columns come from one device GEMM
+        with the sigmoid+normalize fused."""
+        out = self._try_device("predict_proba", X)
+        if out is not None:
+            return out
         probs = np.column_stack(
             [est.predict_proba(X)[:, -1] for est in self.estimators_]
         )
@@ -353,6 +372,12 @@ class DistOneVsOneClassifier(ClassifierMixin, BaseEstimator):
         return self
 
     def predict(self, X):
+        import torch
+
+        if torch.cuda.is_available():
+            fn = self._device_predict_fn("predict", "cuda")
+            if fn is not None:
+                return fn(X)
         k = len(self.classes_)
         n = _n_rows(X)
         votes = np.zeros((n, k))
