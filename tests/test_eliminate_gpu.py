@@ -51,7 +51,10 @@ def test_ovr_device_inference_matches_host():
         LogisticRegression(epochs=12, random_state=0), norm="l1",
         sc=Cluster(require_gpu=True))
     ovr.fit(X, y)
-    host = ovr.predict_proba(X)
+    # explicit host reference (fitted model itself now scores on device)
+    cols = np.column_stack(
+        [est.predict_proba(X)[:, -1] for est in ovr.estimators_])
+    host = cols / cols.sum(axis=1, keepdims=True)
 
     pred = DistPredictor(ovr, sc=None, method="predict_proba")
     dev = pred(X)
