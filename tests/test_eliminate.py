@@ -146,3 +146,25 @@ def test_masked_solve_equals_column_drop():
         Wm[ds_full.intercept_row, 0].item(),
         Wr[ds_red.intercept_row, 0].item(), atol=1e-6)
     assert float(Wm[drop, 0].abs().max()) == 0.0
+
+
+def test_batched_eliminator_multiclass():
+    """Multiclass elimination through the masked batched path: each
+    subset contributes k columns per fold (internal OvR) and the refit
+    model carries the kept-feature coef block per class."""
+    from skdist_amd import Cluster
+    from skdist_amd.models import LogisticRegression
+
+    rng = np.random.default_rng(0)
+    n, f = 3000, 10
+    X = rng.standard_normal((n, f)).astype(np.float32)
+    W = rng.standard_normal((3, 5))
+    y = (X[:, :5] @ W.T).argmax(axis=1)
+    X[:, 5:] = rng.standard_normal((n, 5))
+    el = DistFeatureEliminator(
+        LogisticRegression(epochs=10, random_state=0), sc=Cluster(),
+        min_features_to_select=3, step=2, cv=3)
+    el.fit(X, y)
+    assert set(range(5)) <= set(el.best_features_)
+    assert (el.predict(X) == y).mean() > 0.9
+    assert el.best_estimator_.coef_.shape == (3, len(el.best_features_))
