@@ -65,6 +65,8 @@ def hash_vectorize(docs, n_features=2 ** 20, analyzer="word",
     ext = require_hip()
     mode = {"word": 0, "char_wb": 1}[analyzer]
     min_n, max_n = ngram_range
+    if len(docs) == 0:
+        return sp.csr_matrix((0, n_features), dtype=dtype)
     enc = [
         (d.lower() if lowercase else d).encode("utf-8") for d in docs
     ]
