@@ -42,7 +42,8 @@ sgd.DeviceDataset.shuffled_views = timed(
     "shuffled_views", sgd.DeviceDataset.shuffled_views)
 for mod in (sgd, lin):
     mod.batched_sgd_fit = timed("sgd_fit", mod.batched_sgd_fit)
-    mod.batched_scores = timed("scores", mod.batched_scores)
+    mod.batched_scores_by_fold = timed(
+        "scores", mod.batched_scores_by_fold)
 
 
 def main():
@@ -56,7 +57,7 @@ def main():
 
     def one():
         gs = DistGridSearchCV(
-            LogisticRegression(epochs=20, batch_size=8192, random_state=0),
+            LogisticRegression(epochs=10, batch_size=8192, random_state=0),
             grid, cv=5, scoring="accuracy", sc=cluster)
         gs.fit(X, y)
         return gs
