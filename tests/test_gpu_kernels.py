@@ -255,3 +255,46 @@ def test_sample_weight_hip_matches_eager(monkeypatch):
     # bf16 kernels vs fp32 eager: tolerance like the unweighted tests
     assert (hip.predict(X) == ref.predict(X)).mean() > 0.995
     np.testing.assert_allclose(hip.coef_, ref.coef_, rtol=0.1, atol=0.02)
+
+
+def test_all_device_metrics_reasonable_vs_host():
+    """Every batched device metric (accuracy / roc_auc / f1 /
+    neg_log_loss; r2 / neg_mse via Ridge) agrees with the sc=None host
+    path (sklearn scorers on eager fp32 fits) to solver tolerance."""
+    from skdist_amd import Cluster
+    from skdist_amd.distribute.search import DistGridSearchCV
+    from skdist_amd.models import LogisticRegression, Ridge
+
+    rng = np.random.default_rng(0)
+    n, f = 12000, 16
+    X = rng.standard_normal((n, f)).astype(np.float32)
+    w = rng.standard_normal(f)
+    y = ((X @ w + 0.5 * rng.standard_normal(n)) > 0).astype(np.int64)
+    yr = (X @ w + 0.2 * rng.standard_normal(n)).astype(np.float32)
+
+    for metric in ("accuracy", "roc_auc", "f1", "f1_weighted",
+                   "neg_log_loss"):
+        g_dev = DistGridSearchCV(
+            LogisticRegression(epochs=10, random_state=0),
+            {"C": [1.0]}, cv=3, scoring=metric,
+            sc=Cluster(require_gpu=True))
+        g_dev.fit(X, y)
+        g_host = DistGridSearchCV(
+            LogisticRegression(epochs=10, random_state=0),
+            {"C": [1.0]}, cv=3, scoring=metric, sc=None)
+        g_host.fit(X, y)
+        assert abs(g_dev.best_score_ - g_host.best_score_) < 0.03, (
+            metric, g_dev.best_score_, g_host.best_score_)
+
+    for metric in ("r2", "neg_mean_squared_error"):
+        g_dev = DistGridSearchCV(
+            Ridge(epochs=10, random_state=0), {"alpha": [1.0]},
+            cv=3, scoring=metric, sc=Cluster(require_gpu=True))
+        g_dev.fit(X, yr)
+        g_host = DistGridSearchCV(
+            Ridge(epochs=10, random_state=0), {"alpha": [1.0]},
+            cv=3, scoring=metric, sc=None)
+        g_host.fit(X, yr)
+        rel = abs(g_dev.best_score_ - g_host.best_score_) / max(
+            abs(g_host.best_score_), 1e-6)
+        assert rel < 0.05, (metric, g_dev.best_score_, g_host.best_score_)
