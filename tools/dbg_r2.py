@@ -1,23 +1,27 @@
 import sys
 sys.path.insert(0, "/root/repo")
 import numpy as np
-from skdist_amd import Cluster
-from skdist_amd.distribute.search import DistGridSearchCV
-from skdist_amd.models import Ridge
+from skdist_amd.models._sgd import (ColumnSpec, DeviceDataset,
+                                    batched_sgd_fit,
+                                    batched_scores_by_fold)
 
 rng = np.random.default_rng(0)
 n, f = 12000, 16
 X = rng.standard_normal((n, f)).astype(np.float32)
 w = rng.standard_normal(f)
 yr = (X @ w + 0.2 * rng.standard_normal(n)).astype(np.float32)
-
-for lr, ep in ((0.5, 10), (0.75, 10), (0.5, 20), (0.5, 40)):
-    g = DistGridSearchCV(Ridge(epochs=ep, lr=lr, random_state=0),
-                         {"alpha": [1.0]}, cv=3, scoring="r2",
-                         sc=Cluster(require_gpu=True))
-    g.fit(X, yr)
-    print(f"device search lr={lr} epochs={ep}: r2 {g.best_score_:.5f}")
-g = DistGridSearchCV(Ridge(epochs=10, random_state=0),
-                     {"alpha": [1.0]}, cv=3, scoring="r2", sc=None)
-g.fit(X, yr)
-print("host-path search: r2", round(g.best_score_, 5))
+ds = DeviceDataset(X, yr, device="cuda")
+idx = np.arange(n)
+splits = [(np.setdiff1d(idx, idx[k::3]), idx[k::3]) for k in range(3)]
+ds.set_cv_partition(splits)
+spec = ColumnSpec("cuda", col_fold=np.array([0, 1, 2], dtype=np.int32),
+                  col_class=np.array([-1, -1, -1], dtype=np.int32),
+                  col_lr=np.full(3, 0.5, np.float32),
+                  col_l2=np.full(3, 1.25e-4, np.float32))
+mf = np.array([0, 1, 2])
+cc = np.array([-1, -1, -1], dtype=np.int32)
+for mom in (0.0, 0.9):
+    W = batched_sgd_fit(ds, spec, "squared", 10, 8192, seed=0,
+                        momentum=mom)
+    r = batched_scores_by_fold(ds, W, mf, cc, n_classes=2, metric="r2")
+    print(f"momentum={mom}: r2 {np.round(r, 5)}")
