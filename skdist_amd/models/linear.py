@@ -598,12 +598,18 @@ class LinearSVC(ClassifierMixin, _BatchedLinearBase):
 
 
 class Ridge(RegressorMixin, _BatchedLinearBase):
-    """Batched-SGD ridge regression."""
+    """Batched-SGD ridge regression.
+
+    Default ``momentum=0.0``: unlike the bounded-gradient losses
+    (logistic/hinge, which keep 0.9), heavy momentum at these learning
+    rates oscillates on squared loss and slows convergence several-fold
+    (measured: CV r2 0.89 vs 0.998 at 10 epochs on 12k x 16 synthetic).
+    """
 
     _loss = LOSS_SQUARED
 
     def __init__(self, alpha=1.0, lr=0.5, epochs=20, batch_size=8192,
-                 momentum=0.9, standardize=True, random_state=None, sc=None):
+                 momentum=0.0, standardize=True, random_state=None, sc=None):
         self.alpha = alpha
         self.lr = lr
         self.epochs = epochs
