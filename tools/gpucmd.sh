@@ -1,9 +1,6 @@
 export TMPDIR=/tmp
 mkdir -p /root/repo/gpurun_out
-cd /tmp
-timeout 700 rocprofv3 --kernel-trace --stats -d /root/repo/gpurun_out/prof7 -o prof7 -- python /root/repo/bench.py --steps 3 --warmup 1 > /root/repo/gpurun_out/prof7.log 2>&1
-echo "stats rc=$?"
-timeout 700 rocprofv3 --kernel-trace --pmc SQ_WAVE_CYCLES SQ_WAIT_ANY SQ_ACTIVE_INST_ANY SQ_VALU_MFMA_BUSY_CYCLES -d /root/repo/gpurun_out/pmc7 -o pmc7 -- python /root/repo/bench.py --steps 1 --warmup 1 > /root/repo/gpurun_out/pmc7.log 2>&1
-echo "pmc rc=$?"
 cd /root/repo
-python tools/prof_summary.py gpurun_out/prof7/prof7_results.db 2>&1 | head -8
+echo "== gpu suite ==" && timeout 900 python -m pytest tests -m gpu -q 2>&1 | tail -1
+echo "== smoke ==" && timeout 300 python -c "import __graft_entry__ as g; g.smoke()" 2>&1 | tail -1
+echo "== bench (final) ==" && timeout 700 python bench.py --steps 5 --warmup 1 2>/dev/null | tail -1 | tee gpurun_out/bench_roundend.json | python3 -c "import json,sys; d=json.load(sys.stdin); print(round(d['ms_per_step'],1),'ms/step', round(d['value']),'fits/s', 'vs_baseline', round(d['vs_baseline'],1), 'best', round(d['config']['best_score'],5))"
