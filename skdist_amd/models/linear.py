@@ -89,6 +89,7 @@ class _BatchedLinearBase(BaseEstimator):
         W = batched_sgd_fit(
             ds, spec, self._loss, self.epochs, self.batch_size,
             seed=self._seed(), momentum=self.momentum,
+            lr_decay=getattr(self, "lr_decay", 0.0),
         )
         self._store_fitted(ds, W, is_clf)
         self.n_features_in_ = ds.f
@@ -231,6 +232,7 @@ class _BatchedLinearBase(BaseEstimator):
             W = batched_sgd_fit(
                 ds, spec, self._loss, self.epochs, self.batch_size,
                 seed=self._seed(), momentum=self.momentum,
+                lr_decay=getattr(self, "lr_decay", 0.0),
             )
             fit_time = time.perf_counter() - t0
             t1 = time.perf_counter()
@@ -382,6 +384,7 @@ class _BatchedLinearBase(BaseEstimator):
             W = batched_sgd_fit(
                 ds, spec, self._loss, self.epochs, self.batch_size,
                 seed=self._seed(), momentum=self.momentum,
+                lr_decay=getattr(self, "lr_decay", 0.0),
             )
             scores = batched_scores_by_fold(
                 ds, W, np.asarray(model_folds),
@@ -484,6 +487,7 @@ class _BatchedLinearBase(BaseEstimator):
             W = batched_sgd_fit(
                 ds, spec, self._loss, self.epochs, self.batch_size,
                 seed=self._seed(), momentum=self.momentum,
+                lr_decay=getattr(self, "lr_decay", 0.0),
             )
             per = (time.perf_counter() - t0) / len(ids)
             for ci, pi in enumerate(ids):
@@ -545,8 +549,10 @@ class LogisticRegression(ClassifierMixin, _BatchedLinearBase):
     _loss = LOSS_LOG
 
     def __init__(self, C=1.0, lr=0.5, epochs=20, batch_size=8192,
-                 momentum=0.9, standardize=True, random_state=None, sc=None):
+                 momentum=0.9, lr_decay=0.0, standardize=True,
+                 random_state=None, sc=None):
         self.C = C
+        self.lr_decay = lr_decay
         self.lr = lr
         self.epochs = epochs
         self.batch_size = batch_size
@@ -580,8 +586,10 @@ class LinearSVC(ClassifierMixin, _BatchedLinearBase):
     _loss = LOSS_HINGE
 
     def __init__(self, C=1.0, lr=0.5, epochs=20, batch_size=8192,
-                 momentum=0.9, standardize=True, random_state=None, sc=None):
+                 momentum=0.9, lr_decay=0.0, standardize=True,
+                 random_state=None, sc=None):
         self.C = C
+        self.lr_decay = lr_decay
         self.lr = lr
         self.epochs = epochs
         self.batch_size = batch_size
@@ -609,8 +617,10 @@ class Ridge(RegressorMixin, _BatchedLinearBase):
     _loss = LOSS_SQUARED
 
     def __init__(self, alpha=1.0, lr=0.5, epochs=20, batch_size=8192,
-                 momentum=0.0, standardize=True, random_state=None, sc=None):
+                 momentum=0.0, lr_decay=0.0, standardize=True,
+                 random_state=None, sc=None):
         self.alpha = alpha
+        self.lr_decay = lr_decay
         self.lr = lr
         self.epochs = epochs
         self.batch_size = batch_size

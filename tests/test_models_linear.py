@@ -148,3 +148,16 @@ def test_sample_weight_upweights_rows():
     m0 = LogisticRegression(epochs=15, random_state=0).fit(X, y)
     # weighting toward agreement rows grows |coef| on feature 1
     assert abs(m.coef_[0][1]) > abs(m0.coef_[0][1])
+
+
+def test_lr_decay_param():
+    """lr_decay plumbs through the batched solver (1/(1+decay*epoch))."""
+    rng = np.random.default_rng(4)
+    X = rng.standard_normal((800, 6)).astype(np.float32)
+    y = (X[:, 0] > 0).astype(np.int64)
+    m = LogisticRegression(epochs=15, lr_decay=0.2, random_state=0)
+    m.fit(X, y)
+    assert (m.predict(X) == y).mean() > 0.9
+    m2 = LogisticRegression(epochs=15, random_state=0).fit(X, y)
+    # decay changes the trajectory (not a no-op)
+    assert not np.allclose(m.coef_, m2.coef_)
