@@ -55,3 +55,23 @@ def test_dist_predictor_local_chunked():
     assert (out == model.predict(X)).all()
     proba = DistPredictor(model, method="predict_proba", chunk_rows=64)
     assert np.allclose(proba.predict(X), model.predict_proba(X))
+
+
+def test_prediction_fn_text():
+    """feature_type='text': single raw-text column through a fitted text
+    pipeline (reference predict.py:59-71 'text' marshalling)."""
+    from sklearn.linear_model import LogisticRegression as SkLogReg
+    from sklearn.pipeline import make_pipeline
+
+    from skdist_amd.preprocessing import HashingVectorizerChunked
+
+    docs = np.array(
+        ["good great fine", "bad awful poor", "great nice", "poor bad"]
+        * 25, dtype=object)
+    y = np.array([1, 0, 1, 0] * 25)
+    model = make_pipeline(
+        HashingVectorizerChunked(n_features=4096), SkLogReg())
+    model.fit(docs, y)
+    fn = get_prediction_fn(model, method="predict", feature_type="text")
+    preds = fn(docs[:4])
+    assert list(preds) == [1, 0, 1, 0]
