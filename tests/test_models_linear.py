@@ -161,3 +161,67 @@ def test_lr_decay_param():
     m2 = LogisticRegression(epochs=15, random_state=0).fit(X, y)
     # decay changes the trajectory (not a no-op)
     assert not np.allclose(m.coef_, m2.coef_)
+
+
+def test_fold_mask_excludes_rows_exactly():
+    """A column with col_fold=f must be bit-identical no matter what the
+    fold-f rows' labels are (the mask zeroes their gradients)."""
+    from skdist_amd.models._sgd import (
+        ColumnSpec,
+        DeviceDataset,
+        batched_sgd_fit,
+    )
+
+    rng = np.random.default_rng(5)
+    n, f = 900, 6
+    X = rng.standard_normal((n, f)).astype(np.float32)
+    y = (X[:, 0] > 0).astype(np.int64)
+    idx = np.arange(n)
+    splits = [(np.setdiff1d(idx, idx[k::3]), idx[k::3]) for k in range(3)]
+
+    def solve(yv):
+        ds = DeviceDataset(X, yv, device="cpu")
+        assert ds.set_cv_partition(splits)
+        spec = ColumnSpec(
+            "cpu", col_fold=np.array([0], dtype=np.int32),
+            col_class=np.array([1], dtype=np.int32),
+            col_lr=np.array([0.5], dtype=np.float32),
+            col_l2=np.array([1e-4], dtype=np.float32))
+        return batched_sgd_fit(ds, spec, "log", 6, 256, seed=0).numpy()
+
+    y_flip = y.copy()
+    y_flip[idx[0::3]] = 1 - y_flip[idx[0::3]]  # garbage on fold-0 rows
+    np.testing.assert_array_equal(solve(y), solve(y_flip))
+
+
+def test_ovo_pair_mask_excludes_other_classes_exactly():
+    """A one-vs-one column only sees rows of its two classes: relabeling
+    third-class rows (within the non-pair classes) leaves the pair
+    column bit-identical."""
+    from skdist_amd.models._sgd import (
+        ColumnSpec,
+        DeviceDataset,
+        batched_sgd_fit,
+    )
+
+    rng = np.random.default_rng(6)
+    n, f = 900, 6
+    X = rng.standard_normal((n, f)).astype(np.float32)
+    y = rng.integers(0, 4, size=n)
+
+    def solve(yv):
+        ds = DeviceDataset(X, yv, device="cpu",
+                           classes=np.array([0, 1, 2, 3]))
+        ds.set_cv_partition([])
+        spec = ColumnSpec(
+            "cpu", col_fold=np.array([-2], dtype=np.int32),
+            col_class=np.array([1], dtype=np.int32),
+            col_lr=np.array([0.5], dtype=np.float32),
+            col_l2=np.array([1e-4], dtype=np.float32),
+            col_class2=np.array([0], dtype=np.int32))
+        return batched_sgd_fit(ds, spec, "log", 6, 256, seed=0).numpy()
+
+    y_swap = y.copy()
+    mask = y >= 2
+    y_swap[mask] = 5 - y_swap[mask]  # 2<->3, stays outside the (0,1) pair
+    np.testing.assert_array_equal(solve(y), solve(y_swap))
