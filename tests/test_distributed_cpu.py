@@ -228,7 +228,7 @@ def _run_spmd(scenario, world_size=2):
         p.start()
     outs = {}
     for _ in range(world_size):
-        rank, status, payload = q.get(timeout=240)
+        rank, status, payload = q.get(timeout=840)
         assert status == "ok", f"rank {rank}: {payload}"
         outs[rank] = payload
     for p in procs:
@@ -236,7 +236,7 @@ def _run_spmd(scenario, world_size=2):
     return outs
 
 
-@pytest.mark.timeout(300)
+@pytest.mark.timeout(900)
 def test_spmd_batched_gloo():
     outs = _run_spmd("batched")
     assert outs[0]["best_score"] > 0.98
@@ -244,14 +244,14 @@ def test_spmd_batched_gloo():
     assert np.allclose(outs[0]["scores"], outs[1]["scores"])
 
 
-@pytest.mark.timeout(300)
+@pytest.mark.timeout(900)
 def test_spmd_generic_gloo():
     outs = _run_spmd("generic")
     assert outs[0]["best_score"] > 0.9
     assert np.allclose(outs[0]["scores"], outs[1]["scores"])
 
 
-@pytest.mark.timeout(300)
+@pytest.mark.timeout(900)
 def test_spmd_forest_gloo():
     outs = _run_spmd("forest")
     assert outs[0]["n_trees"] == 12
@@ -260,7 +260,7 @@ def test_spmd_forest_gloo():
     assert np.isclose(outs[0]["proba_sum"], outs[1]["proba_sum"])
 
 
-@pytest.mark.timeout(300)
+@pytest.mark.timeout(900)
 def test_spmd_ovr_gloo():
     outs = _run_spmd("ovr")
     assert outs[0]["k"] == 4
@@ -268,7 +268,7 @@ def test_spmd_ovr_gloo():
     assert np.allclose(outs[0]["proba0"], outs[1]["proba0"])
 
 
-@pytest.mark.timeout(300)
+@pytest.mark.timeout(900)
 def test_spmd_eliminate_gloo():
     outs = _run_spmd("eliminate")
     assert outs[0]["kept"] == outs[1]["kept"]
@@ -277,14 +277,14 @@ def test_spmd_eliminate_gloo():
     assert np.allclose(outs[0]["scores"], outs[1]["scores"])
 
 
-@pytest.mark.timeout(300)
+@pytest.mark.timeout(900)
 def test_spmd_predictor_gloo():
     outs = _run_spmd("predictor")
     assert outs[0]["match"] and outs[1]["match"]
     assert outs[0]["n"] == 2000
 
 
-@pytest.mark.timeout(300)
+@pytest.mark.timeout(900)
 def test_spmd_encoder_gloo():
     outs = _run_spmd("encoder")
     assert outs[0]["shape"] == outs[1]["shape"]
