@@ -211,8 +211,8 @@ class DistOneVsRestClassifier(ClassifierMixin, BaseEstimator):
         fused link/normalize) when a device is visible; None -> host."""
         import torch
 
-        if not torch.cuda.is_available():
-            return None
+        if sp.issparse(X) or not torch.cuda.is_available():
+            return None  # sparse X scores through the host estimators
         fn = self._device_predict_fn(method, "cuda")
         return None if fn is None else fn(X)
 
@@ -374,7 +374,7 @@ class DistOneVsOneClassifier(ClassifierMixin, BaseEstimator):
     def predict(self, X):
         import torch
 
-        if torch.cuda.is_available():
+        if not sp.issparse(X) and torch.cuda.is_available():
             fn = self._device_predict_fn("predict", "cuda")
             if fn is not None:
                 return fn(X)

@@ -60,6 +60,32 @@ def _use_hip(device):
     return True
 
 
+def as_dense_f32(X):
+    """Dense float32 array from X; scipy-sparse inputs are densified.
+
+    The batched solver is a dense-MFMA design:
+    288 GB of HBM3E per GPU makes densifying hashed/sparse feature
+    matrices the right MI355X-first call.  A size guard raises a clear
+    error instead of an opaque allocation failure when the dense form
+    would be unreasonable (override with SKDIST_AMD_DENSIFY_GB).
+    """
+    import scipy.sparse as sp
+
+    if sp.issparse(X):
+        gb = X.shape[0] * X.shape[1] * 4.0 / 1e9
+        limit = float(os.environ.get("SKDIST_AMD_DENSIFY_GB", "64"))
+        if gb > limit:
+            raise ValueError(
+                f"sparse X would densify to {gb:.1f} GB (> {limit:.0f} GB; "
+                "set SKDIST_AMD_DENSIFY_GB to raise the limit). The "
+                "batched linear solver is dense — for wider sparse "
+                "problems use a host sklearn estimator, which rides the "
+                "generic task fan-out."
+            )
+        return np.ascontiguousarray(X.toarray(), dtype=np.float32)
+    return np.asarray(X, dtype=np.float32)
+
+
 class DeviceDataset:
     """(X, y) resident on one device, shared by every fit in a search.
 
@@ -73,6 +99,8 @@ class DeviceDataset:
     def __init__(self, X, y, cluster=None, device=None, standardize=True,
                  classes=None, sample_weight=None):
         self.cluster = cluster
+        if X is not None:
+            X = as_dense_f32(X)
         if device is None:
             device = cluster.device if cluster is not None else (
                 torch.device("cuda") if torch.cuda.is_available()

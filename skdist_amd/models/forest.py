@@ -112,6 +112,10 @@ class HistTree:
 
     def apply(self, X):
         """Leaf NODE index per row (vectorized host traversal)."""
+        import scipy.sparse as sp
+
+        if sp.issparse(X):
+            X = X.toarray()
         X = np.asarray(X, dtype=np.float32)
         node = np.zeros(len(X), dtype=np.int64)
         active = self.feature[node] >= 0
@@ -997,6 +1001,10 @@ class FlatForest:
 
     def predict_value(self, X, chunk_rows=1 << 22):
         """Mean leaf payload over trees: [rows, vs] numpy."""
+        import scipy.sparse as sp
+
+        if sp.issparse(X):
+            X = X.toarray()
         X = np.ascontiguousarray(X, dtype=np.float32)
         outs = []
         for lo in range(0, len(X), chunk_rows):
@@ -1022,6 +1030,10 @@ class FlatForest:
     def apply(self, X, chunk_rows=1 << 21):
         """Leaf node id per (row, tree): [rows, n_trees] int32 — ids are
         tree-local (matching HistTree.apply) for embedding parity."""
+        import scipy.sparse as sp
+
+        if sp.issparse(X):
+            X = X.toarray()
         X = np.ascontiguousarray(X, dtype=np.float32)
         roots_np = self.roots.cpu().numpy().astype(np.int64)
         outs = []

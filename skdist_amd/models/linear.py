@@ -61,7 +61,7 @@ class _BatchedLinearBase(BaseEstimator):
         sc = getattr(self, "sc", None)
         is_clf = isinstance(self, ClassifierMixin)
         ds = DeviceDataset(
-            np.asarray(X, dtype=np.float32), np.asarray(y),
+            X, np.asarray(y),
             cluster=None,  # single fit: local device, no broadcast
             device=sc.device if sc is not None else None,
             standardize=self.standardize,
@@ -121,8 +121,16 @@ class _BatchedLinearBase(BaseEstimator):
     # host-side inference (numpy, sklearn-compatible)
     # ------------------------------------------------------------------ #
     def decision_function(self, X):
-        X = np.asarray(X, dtype=np.float64)
-        scores = X @ self.coef_.T + self.intercept_
+        import scipy.sparse as _sp
+
+        if _sp.issparse(X):
+            scores = (
+                np.asarray(X.astype(np.float64) @ self.coef_.T)
+                + self.intercept_
+            )
+        else:
+            X = np.asarray(X, dtype=np.float64)
+            scores = X @ self.coef_.T + self.intercept_
         if scores.ndim == 2 and scores.shape[1] == 1:
             return scores.ravel()
         return scores
@@ -153,7 +161,7 @@ class _BatchedLinearBase(BaseEstimator):
         is_clf = isinstance(self, ClassifierMixin)
         t0 = time.perf_counter()
         ds = DeviceDataset(
-            None if X is None else np.asarray(X, dtype=np.float32),
+            X,
             None if y is None else np.asarray(y),
             cluster=cluster, standardize=self.standardize,
             sample_weight=sample_weight,
@@ -318,7 +326,7 @@ class _BatchedLinearBase(BaseEstimator):
         is_clf = isinstance(self, ClassifierMixin)
         t0 = time.perf_counter()
         ds = DeviceDataset(
-            None if X is None else np.asarray(X, dtype=np.float32),
+            X,
             None if y is None else np.asarray(y),
             cluster=cluster, standardize=self.standardize,
         )
@@ -448,7 +456,7 @@ class _BatchedLinearBase(BaseEstimator):
         """
         t0 = time.perf_counter()
         ds = DeviceDataset(
-            None if X is None else np.asarray(X, dtype=np.float32),
+            X,
             None if y is None else np.asarray(y),
             cluster=cluster, standardize=self.standardize,
         )
