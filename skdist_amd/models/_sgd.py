@@ -97,7 +97,7 @@ class DeviceDataset:
     """
 
     def __init__(self, X, y, cluster=None, device=None, standardize=True,
-                 classes=None, sample_weight=None):
+                 classes=None, sample_weight=None, task=None):
         self.cluster = cluster
         if X is not None:
             X = as_dense_f32(X)
@@ -112,42 +112,56 @@ class DeviceDataset:
         )
         self.comp_dtype = comp_dtype
 
+        # labels normalize on the host BEFORE any tensor/broadcast work:
+        # string/object classes can't tensorize, and encoding once on the
+        # data-owning rank beats re-encoding on every rank.
+        kind = cls_arr = y_host = None
+        if y is not None:
+            y_np = np.asarray(y)
+            if classes is not None:
+                cls_arr = np.asarray(classes)
+                y_host = np.ascontiguousarray(
+                    np.searchsorted(cls_arr, y_np), dtype=np.int32
+                )
+                kind = "cls"
+            elif task == "reg" or (task is None and y_np.dtype.kind == "f"):
+                # regression targets keep their VALUES (a regressor given
+                # integer y must not regress on label-encoded indices)
+                y_host = np.ascontiguousarray(y_np, dtype=np.float32)
+                kind = "reg"
+            elif task == "cls" or y_np.dtype.kind != "f":
+                cls_arr, enc = np.unique(y_np, return_inverse=True)
+                y_host = np.ascontiguousarray(enc, dtype=np.int32)
+                kind = "cls"
+
         if cluster is not None and cluster.distributed:
             Xt = None
             yt = None
             if X is not None:
                 Xt = torch.as_tensor(np.ascontiguousarray(X), dtype=torch.float32)
-            if y is not None:
-                yt = torch.as_tensor(np.ascontiguousarray(y))
+            if y_host is not None:
+                yt = torch.as_tensor(y_host)
             Xt = cluster.bcast_tensor(
                 Xt.to(cluster.device) if Xt is not None else None
             )
             yt = cluster.bcast_tensor(
                 yt.to(cluster.device) if yt is not None else None
             )
+            kind, cls_arr = cluster.bcast_obj((kind, cls_arr))
         else:
             Xt = torch.as_tensor(
                 np.ascontiguousarray(X), dtype=torch.float32
             ).to(self.device)
-            yt = torch.as_tensor(np.ascontiguousarray(y)).to(self.device)
+            yt = torch.as_tensor(y_host).to(self.device)
 
         self.n, self.f = Xt.shape
-        # label encoding (host-visible classes, device int targets)
-        if yt.dtype.is_floating_point and classes is None:
-            # regression targets
+        if kind == "reg":
             self.classes_ = None
             self.y_float = yt.to(torch.float32)
             self.y_int = None
         else:
-            y_np = yt.cpu().numpy()
-            if classes is not None:
-                self.classes_ = np.asarray(classes)
-                enc = np.searchsorted(self.classes_, y_np)
-            else:
-                self.classes_, enc = np.unique(y_np, return_inverse=True)
-            self.y_int = torch.as_tensor(
-                np.ascontiguousarray(enc, dtype=np.int32), device=self.device
-            )
+            self.classes_ = cls_arr
+            self.y_int = yt.to(torch.int32)
             self.y_float = self.y_int.to(torch.float32)
 
         # standardize + augment with ones column (intercept)

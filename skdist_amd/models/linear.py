@@ -61,11 +61,12 @@ class _BatchedLinearBase(BaseEstimator):
         sc = getattr(self, "sc", None)
         is_clf = isinstance(self, ClassifierMixin)
         ds = DeviceDataset(
-            X, np.asarray(y),
+            X, y,
             cluster=None,  # single fit: local device, no broadcast
             device=sc.device if sc is not None else None,
             standardize=self.standardize,
             sample_weight=sample_weight,
+            task="cls" if is_clf else "reg",
         )
         ds.set_cv_partition([])  # no CV mask
         if is_clf:
@@ -161,10 +162,10 @@ class _BatchedLinearBase(BaseEstimator):
         is_clf = isinstance(self, ClassifierMixin)
         t0 = time.perf_counter()
         ds = DeviceDataset(
-            X,
-            None if y is None else np.asarray(y),
+            X, y,
             cluster=cluster, standardize=self.standardize,
             sample_weight=sample_weight,
+            task="cls" if is_clf else "reg",
         )
         if not ds.set_cv_partition(cv_splits):
             raise FallbackToGeneric("cv splits do not partition the data")
@@ -326,9 +327,9 @@ class _BatchedLinearBase(BaseEstimator):
         is_clf = isinstance(self, ClassifierMixin)
         t0 = time.perf_counter()
         ds = DeviceDataset(
-            X,
-            None if y is None else np.asarray(y),
+            X, y,
             cluster=cluster, standardize=self.standardize,
+            task="cls" if is_clf else "reg",
         )
         if not ds.set_cv_partition(cv_splits):
             raise FallbackToGeneric("cv splits do not partition the data")
@@ -456,8 +457,7 @@ class _BatchedLinearBase(BaseEstimator):
         """
         t0 = time.perf_counter()
         ds = DeviceDataset(
-            X,
-            None if y is None else np.asarray(y),
+            X, y,
             cluster=cluster, standardize=self.standardize,
         )
         ds.set_cv_partition([])
