@@ -129,6 +129,9 @@ class Encoderizer(TransformerMixin, BaseEstimator):
             out = X
         elif isinstance(X, dict):
             out = pd.DataFrame.from_dict(X, orient="columns")
+        elif isinstance(X, list) and X and isinstance(X[0], dict):
+            # list of record dicts → one column per key
+            out = pd.DataFrame.from_records(X)
         elif isinstance(X, (np.ndarray, list)):
             if fit:
                 if self.col_names is None:

@@ -133,3 +133,13 @@ def test_transformer_weights(mixed_df24):
     a = enc.transform(mixed_df)
     b = enc_w.transform(mixed_df)
     assert abs(b.sum() - a.sum()) > 0  # weighting changed something
+
+
+def test_list_of_record_dicts():
+    """Record-style input coerces to a frame (reference
+    encoder.py:237-266 accepts pandas/dict/numpy/list)."""
+    recs = [{"txt": f"doc {i % 3}", "num": float(i)} for i in range(40)]
+    enc = Encoderizer(size="small")
+    T = enc.fit_transform(recs)
+    assert T.shape[0] == 40
+    assert enc.transform(recs[:5]).shape == (5, T.shape[1])
