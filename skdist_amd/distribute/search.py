@@ -492,7 +492,7 @@ class DistMultiModelSearch(BaseEstimator, MetaEstimatorMixin):
                     continue
                 try:
                     out = est.batched_cv_fit_score(
-                        np.asarray(X), None if y is None else np.asarray(y),
+                        X, y,
                         candidate_params=cand, cv_splits=folds,
                         scoring=self.scoring, scorers=None, cluster=sc,
                     )
