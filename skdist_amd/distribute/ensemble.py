@@ -123,7 +123,10 @@ class DistBaseForest(BaseEstimator):
                 f"n_estimators={self.n_estimators} must be >= "
                 f"len(estimators_)={len(self.estimators_)} when warm_start"
             )
-        seeds = seeds[:n_more]
+        # warm_start: the seed stream is redrawn deterministically, so skip
+        # the seeds already consumed — added trees must NOT repeat the
+        # existing trees' seeds (they would be exact duplicates)
+        seeds = seeds[len(self.estimators_):]
 
         if self._device_fit_ok(sc, X, sample_weight):
             results = self._fit_trees_device(sc, X, y, seeds)

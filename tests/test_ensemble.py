@@ -168,3 +168,20 @@ def test_warm_start_adds_trees():
     clf.n_estimators = 7
     clf.fit(X, y)
     assert len(clf.estimators_) == 7
+
+
+def test_warm_start_new_trees_get_fresh_seeds():
+    """Warm-started additions must not duplicate existing trees
+    (regression: the redrawn seed stream was sliced from the front)."""
+    rng = np.random.default_rng(0)
+    X = rng.standard_normal((200, 5)).astype(np.float32)
+    y = (X[:, 0] > 0).astype(int)
+    rf = DistRandomForestClassifier(
+        n_estimators=3, random_state=0, warm_start=True
+    ).fit(X, y)
+    first = list(rf._seeds)
+    rf.set_params(n_estimators=6)
+    rf.fit(X, y)
+    second = list(rf._seeds)
+    assert len(rf.estimators_) == 6
+    assert not (set(first) & set(second))
