@@ -206,6 +206,12 @@ class DistOneVsRestClassifier(ClassifierMixin, BaseEstimator):
                 cols.append(est.decision_function(X))
         return np.column_stack(cols)
 
+    def _binary_single_column(self):
+        """Binary y on the generic path: LabelBinarizer yields ONE column
+        (positive class = classes_[1]); argmax over it would always pick
+        class 0, so the single-column case thresholds instead."""
+        return len(self.classes_) == 2 and len(self.estimators_) == 1
+
     def _try_device(self, method, X):
         """Opportunistic GPU scoring of the fitted model (one GEMM +
         fused link/normalize) when a device is visible; None -> host."""
@@ -213,6 +219,8 @@ class DistOneVsRestClassifier(ClassifierMixin, BaseEstimator):
 
         if sp.issparse(X) or not torch.cuda.is_available():
             return None  # sparse X scores through the host estimators
+        if self._binary_single_column():
+            return None  # thresholded host path (argmax semantics differ)
         fn = self._device_predict_fn(method, "cuda")
         return None if fn is None else fn(X)
 
@@ -227,6 +235,11 @@ class DistOneVsRestClassifier(ClassifierMixin, BaseEstimator):
             if self.mlb_ is not None:
                 return self.mlb_.inverse_transform(ind)
             return ind
+        if self._binary_single_column():
+            thr = (
+                0.5 if hasattr(self.estimators_[0], "predict_proba") else 0.0
+            )
+            return self.classes_[(scores[:, 0] > thr).astype(int)]
         return self.classes_[scores.argmax(axis=1)]
 
     def predict_proba(self, X):
@@ -239,12 +252,17 @@ class DistOneVsRestClassifier(ClassifierMixin, BaseEstimator):
         probs = np.column_stack(
             [est.predict_proba(X)[:, -1] for est in self.estimators_]
         )
+        if self._binary_single_column():
+            probs = np.column_stack([1.0 - probs[:, 0], probs[:, 0]])
         if self.norm:
             return normalize(probs, norm=self.norm)
         return probs
 
     def decision_function(self, X):
-        return self._scores(X)
+        scores = self._scores(X)
+        if self._binary_single_column():
+            return scores.ravel()
+        return scores
 
     # ------------------------------------------------------------------ #
     # batched device inference (DistPredictor hook): all k binary models

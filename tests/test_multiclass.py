@@ -132,3 +132,26 @@ def test_ovr_multilabel_sequences():
     preds = clf.predict(X[:3])
     assert set(preds[0]) == {"a"}
     assert set(preds[2]) == {"a", "b"}
+
+
+def test_ovr_binary_y():
+    """Binary y through OvR: single LabelBinarizer column must threshold,
+    not argmax (regression: predictions collapsed to the first class);
+    matches sklearn's OneVsRestClassifier exactly."""
+    from sklearn.linear_model import LogisticRegression as SkLR
+    from sklearn.multiclass import OneVsRestClassifier
+
+    rng = np.random.default_rng(0)
+    X = rng.standard_normal((200, 6)).astype(np.float32)
+    y = (X[:, 0] > 0).astype(int)
+    ours = DistOneVsRestClassifier(SkLR(solver="liblinear")).fit(X, y)
+    ref = OneVsRestClassifier(SkLR(solver="liblinear")).fit(X, y)
+    assert (ours.predict(X) == ref.predict(X)).all()
+    assert ours.predict_proba(X).shape == (200, 2)
+    assert np.allclose(ours.predict_proba(X).sum(axis=1), 1.0)
+    assert ours.decision_function(X).shape == (200,)
+    # native estimator too
+    nat = DistOneVsRestClassifier(
+        LogisticRegression(epochs=10, random_state=0)
+    ).fit(X, y)
+    assert (nat.predict(X) == y).mean() > 0.9
