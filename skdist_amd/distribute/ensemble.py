@@ -75,6 +75,50 @@ def _fit_one_tree(tree_proto, X, y, seed, bootstrap, sample_weight,
     return tree
 
 
+def get_single_oof(clf, X, y, train_index, test_index):
+    """One out-of-fold fit+predict_proba (reference ensemble.py:112-128):
+    fit on ``train_index`` rows, return (test_index, proba on them)."""
+    from .base import _clone
+    from .utils import _safe_split
+
+    est = _clone(clf)
+    if hasattr(est, "sc"):
+        est.sc = None
+    X_tr, y_tr = _safe_split(est, X, y, train_index)
+    X_te, _ = _safe_split(est, X, y, test_index, train_index)
+    est.fit(X_tr, y_tr)
+    return test_index, est.predict_proba(X_te)
+
+
+def get_oof(clf, X, y, n_splits=5, sc=None):
+    """Out-of-fold probabilities + a final full fit (reference
+    ensemble.py:130-151).  The per-fold fits are independent tasks, so
+    with a Cluster they shard across ranks like any other fan-out."""
+    from sklearn.model_selection import KFold
+
+    y = np.asarray(y)
+    folds = list(KFold(n_splits=n_splits).split(X))
+    oof = np.zeros((y.shape[0], len(np.unique(y))))
+
+    def task_fn(task):
+        train_idx, test_idx = task
+        return get_single_oof(clf, X, y, train_idx, test_idx)
+
+    if sc is None:
+        results = run_local_tasks(task_fn, folds)
+    else:
+        results = sc.run_tasks(task_fn, folds)
+    for test_idx, proba in results:
+        oof[np.asarray(test_idx)] = proba
+    from .base import _clone
+
+    fitted = _clone(clf)
+    if hasattr(fitted, "sc"):
+        fitted.sc = None
+    fitted.fit(X, y)
+    return fitted, oof
+
+
 class DistBaseForest(BaseEstimator):
     """Shared fan-out machinery for all forest classes
     (reference ensemble.py:154-340).

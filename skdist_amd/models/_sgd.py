@@ -302,8 +302,9 @@ def batched_sgd_fit(ds, spec, loss, epochs, batch_size, seed=0,
                     momentum=0.0, lr_decay=0.0, force_eager=False):
     """Train all columns; returns W [f+1, ncols] fp32 on ds.device.
 
-    Mini-batches walk a fixed seeded permutation of the rows (re-drawn
-    per epoch on the host RNG, applied as a device gather).
+    Mini-batches walk ONE fixed seeded permutation of the rows (drawn on
+    the host RNG, applied as a device gather); batch composition stays
+    fixed across epochs, matching the HIP path (hip_sgd_solve).
     """
     device = ds.device
     n, fa = ds.Xaug.shape

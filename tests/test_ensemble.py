@@ -185,3 +185,24 @@ def test_warm_start_new_trees_get_fresh_seeds():
     second = list(rf._seeds)
     assert len(rf.estimators_) == 6
     assert not (set(first) & set(second))
+
+
+def test_get_oof_helpers():
+    """Out-of-fold helpers (reference ensemble.py:112-151): oof rows come
+    from fits that never saw them; the returned model is fully fitted."""
+    from skdist_amd.distribute.ensemble import get_oof, get_single_oof
+
+    rng = np.random.default_rng(0)
+    X = rng.standard_normal((300, 6)).astype(np.float32)
+    y = (X[:, 0] > 0).astype(int)
+    clf = DistRandomForestClassifier(n_estimators=10, random_state=0)
+    fitted, oof = get_oof(clf, X, y, n_splits=3)
+    assert oof.shape == (300, 2)
+    assert np.allclose(oof.sum(axis=1), 1.0)
+    assert len(fitted.estimators_) == 10
+    # oof probabilities are predictive (out-of-sample accuracy)
+    assert ((oof[:, 1] > 0.5).astype(int) == y).mean() > 0.8
+    idx_tr = np.arange(150)
+    idx_te = np.arange(150, 300)
+    te, proba = get_single_oof(clf, X, y, idx_tr, idx_te)
+    assert (te == idx_te).all() and proba.shape == (150, 2)
