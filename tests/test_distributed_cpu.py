@@ -242,6 +242,30 @@ def test_spmd_batched_gloo():
     assert outs[0]["best_score"] > 0.98
     # every rank assembled identical results
     assert np.allclose(outs[0]["scores"], outs[1]["scores"])
+    # determinism across WORLD SIZES: the same batched solve at
+    # world_size=1 (in this process) must reproduce the 2-rank run
+    # exactly — candidate sharding must not change any column's result
+    # (results keyed by task id, seeded solver walk — SURVEY.md §5).
+    # (The sc=None generic path is a different execution plan — per-fold
+    # standardization and per-task fits — and matches only to solver
+    # tolerance, not bitwise.)
+    from sklearn.datasets import load_breast_cancer
+
+    from skdist_amd import Cluster
+    from skdist_amd.distribute.search import DistGridSearchCV
+    from skdist_amd.models import LogisticRegression
+
+    X, y = load_breast_cancer(return_X_y=True)
+    gs = DistGridSearchCV(
+        LogisticRegression(epochs=10, random_state=0),
+        {"C": [0.01, 0.1, 1.0, 10.0]},
+        cv=3, scoring="roc_auc", sc=Cluster(),
+    )
+    gs.fit(X, y)
+    assert np.allclose(
+        outs[0]["scores"], list(gs.cv_results_["mean_test_score"]),
+        rtol=0, atol=1e-12,
+    )
 
 
 @pytest.mark.timeout(900)
