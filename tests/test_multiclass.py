@@ -155,3 +155,22 @@ def test_ovr_binary_y():
         LogisticRegression(epochs=10, random_state=0)
     ).fit(X, y)
     assert (nat.predict(X) == y).mean() > 0.9
+
+
+def test_negatives_mask_modes():
+    """ratio / multiplier / absolute cap semantics
+    (reference multiclass.py:76-106)."""
+    from skdist_amd.distribute.multiclass import _negatives_mask
+
+    rng = np.random.default_rng(0)
+    y = (rng.random(2000) < 0.05).astype(int)
+    n_pos = int(y.sum())
+    n_neg = 2000 - n_pos
+    m = _negatives_mask(y, 0.5, method="ratio", random_state=0)
+    assert m.sum() == n_pos + int(0.5 * n_neg)
+    m = _negatives_mask(y, 3.0, method="multiplier", random_state=0)
+    assert m.sum() == n_pos + 3 * n_pos
+    m = _negatives_mask(y, 100, random_state=0)
+    assert m.sum() == n_pos + 100
+    # all positives always kept
+    assert m[y == 1].all()
