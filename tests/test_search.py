@@ -206,3 +206,26 @@ def test_search_with_sample_weight_batched(small_xy):
     # fitted models must be identical: zero-weight rows have no gradient
     np.testing.assert_allclose(
         g1.best_estimator_.coef_, g2.best_estimator_.coef_, atol=1e-7)
+
+
+def test_fit_params_passthrough_host_estimator():
+    """Arbitrary estimator-specific fit kwargs travel through the task
+    fan-out to every worker fit (reference test_search.py:86-101 did
+    this with xgboost eval_set/early_stopping)."""
+    from sklearn.ensemble import GradientBoostingClassifier
+
+    X = np.array([[1, 1, 1], [0, 0, 0], [-1, -1, -1]] * 60, dtype=float)
+    y = np.array([0, 0, 1] * 60)
+    seen = []
+
+    def monitor(i, est, locals_):  # sklearn GBT's fit callback kwarg
+        seen.append(i)
+        return False
+
+    gs = DistRandomizedSearchCV(
+        GradientBoostingClassifier(n_estimators=5, random_state=0),
+        {"max_depth": [2, 3]}, cv=3, n_iter=2, random_state=0,
+    )
+    gs.fit(X, y, monitor=monitor)
+    assert np.allclose(gs.predict(X[:3]), [0, 0, 1])
+    assert len(seen) > 0  # the kwarg reached the worker fits
