@@ -10,6 +10,13 @@ builder.  All of them keep the sklearn API and pickle with host numpy
 weights only.
 """
 
+from .boosting import (
+    HistGradientBoostingClassifier,
+    HistGradientBoostingRegressor,
+)
 from .linear import LinearSVC, LogisticRegression, Ridge
 
-__all__ = ["LogisticRegression", "LinearSVC", "Ridge"]
+__all__ = [
+    "LogisticRegression", "LinearSVC", "Ridge",
+    "HistGradientBoostingClassifier", "HistGradientBoostingRegressor",
+]
