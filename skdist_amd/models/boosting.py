@@ -47,7 +47,9 @@ class _BaseHistGB(BaseEstimator):
 
     def __init__(self, n_estimators=100, learning_rate=0.1, max_depth=3,
                  subsample=1.0, min_samples_split=2, min_samples_leaf=1,
-                 max_features=None, random_state=None, sc=None):
+                 max_features=None, n_iter_no_change=None,
+                 validation_fraction=0.1, tol=1e-4, random_state=None,
+                 sc=None):
         self.n_estimators = n_estimators
         self.learning_rate = learning_rate
         self.max_depth = max_depth
@@ -55,6 +57,9 @@ class _BaseHistGB(BaseEstimator):
         self.min_samples_split = min_samples_split
         self.min_samples_leaf = min_samples_leaf
         self.max_features = max_features
+        self.n_iter_no_change = n_iter_no_change
+        self.validation_fraction = validation_fraction
+        self.tol = tol
         self.random_state = random_state
         self.sc = sc
 
@@ -88,6 +93,21 @@ class _BaseHistGB(BaseEstimator):
             bootstrap=False,
         )
 
+        # early stopping: hold out a validation slice whose rows train no
+        # tree (weight 0 through the builder's weight plane) and track
+        # its loss (sklearn's n_iter_no_change semantics)
+        train_mask = None
+        val_idx = None
+        if self.n_iter_no_change is not None:
+            vn = max(1, int(round(self.validation_fraction * n)))
+            if vn >= n:
+                raise ValueError("validation_fraction leaves no train rows")
+            val_idx = rng.permutation(n)[:vn]
+            train_mask = np.ones(n, dtype=np.float32)
+            train_mask[val_idx] = 0.0
+        best_loss = np.inf
+        stall = 0
+
         stages = []
         for _ in range(self.n_estimators):
             if self.subsample < 1.0:
@@ -98,6 +118,8 @@ class _BaseHistGB(BaseEstimator):
                     mask[:] = 1.0
             else:
                 mask = None
+            if train_mask is not None:
+                mask = train_mask if mask is None else mask * train_mask
             grad, hess = self._gradients(y, F)   # [n, K] each
             round_trees = []
             for k in range(K):
@@ -113,10 +135,20 @@ class _BaseHistGB(BaseEstimator):
                 F[:, k] += self.learning_rate * tree.value[rows, 0]
                 round_trees.append(tree)
             stages.append(round_trees)
+            if val_idx is not None:
+                loss = self._loss(val_idx, F)
+                if loss < best_loss - self.tol:
+                    best_loss = loss
+                    stall = 0
+                else:
+                    stall += 1
+                    if stall >= self.n_iter_no_change:
+                        break
 
         self.stages_ = stages
         self.n_features_in_ = X.shape[1]
-        for a in ("_enc", "_y"):  # training labels don't belong in the pickle
+        self.n_estimators_ = len(stages)
+        for a in ("_enc", "_yv"):  # training labels don't belong in the pickle
             if hasattr(self, a):
                 delattr(self, a)
         _strip(self)
@@ -228,8 +260,13 @@ class HistGradientBoostingRegressor(RegressorMixin, _BaseHistGB):
     """Least-squares gradient boosting on binned HistTrees."""
 
     def _init_raw(self, y, n):
-        self.init_raw_ = float(np.asarray(y, dtype=np.float64).mean())
+        self._yv = np.asarray(y, dtype=np.float64)
+        self.init_raw_ = float(self._yv.mean())
         return np.full((n, 1), self.init_raw_), 1
+
+    def _loss(self, idx, F):
+        d = self._yv[idx] - F[idx, 0]
+        return float(np.mean(d * d))
 
     def _base_raw(self):
         return np.array([self.init_raw_])
@@ -283,6 +320,17 @@ class HistGradientBoostingClassifier(ClassifierMixin, _BaseHistGB):
 
     def _newton_scale(self, K):
         return 1.0 if K == 1 else (K - 1.0) / K
+
+    def _loss(self, idx, F):
+        # mean deviance (neg log-likelihood) on the validation slice
+        if F.shape[1] == 1:
+            z = F[idx, 0]
+            t = self._enc[idx]
+            return float(np.mean(np.logaddexp(0.0, z) - t * z))
+        z = F[idx]
+        lse = np.log(np.exp(z - z.max(axis=1, keepdims=True)).sum(axis=1))
+        lse += z.max(axis=1)
+        return float(np.mean(lse - z[np.arange(len(idx)), self._enc[idx]]))
 
     def decision_function(self, X):
         r = self._raw_scores(X)

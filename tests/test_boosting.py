@@ -133,3 +133,17 @@ def test_rides_search_and_predictor(data):
     pred = DistPredictor(blob.best_estimator_, method="predict_proba")
     out = pred(X)
     assert np.allclose(out, blob.best_estimator_.predict_proba(X))
+
+
+def test_early_stopping(data):
+    X, t = data
+    y = (t > np.median(t)).astype(int)
+    m = HistGradientBoostingClassifier(
+        n_estimators=200, n_iter_no_change=5, random_state=0
+    ).fit(X, y)
+    assert m.n_estimators_ < 200
+    assert (m.predict(X) == y).mean() > 0.9
+    r = HistGradientBoostingRegressor(
+        n_estimators=300, n_iter_no_change=5, random_state=0
+    ).fit(X, X[:, 0] * 2.0)
+    assert r.n_estimators_ < 300
