@@ -36,6 +36,23 @@ def _want_distributed():
     return "RANK" in os.environ and "WORLD_SIZE" in os.environ
 
 
+class TaskFailedError(RuntimeError):
+    """A sharded task raised on its worker rank; re-raised identically on
+    EVERY rank after the result gather (no rank is left hanging in the
+    collective — reference analog: Spark surfaces task failures to the
+    driver, SURVEY.md §5 'failure detection')."""
+
+
+class _TaskError:
+    """Picklable failure record that travels through the result gather."""
+
+    def __init__(self, task_id, exc_type, msg, tb):
+        self.task_id = task_id
+        self.exc_type = exc_type
+        self.msg = msg
+        self.tb = tb
+
+
 class Cluster:
     """Handle to the single-node GPU scheduler (the ``sc=`` argument).
 
@@ -164,10 +181,36 @@ class Cluster:
         on this rank's shard only; cross-rank results travel as pickled
         objects (they are scores or compact fitted-model blobs — the data
         itself never moves after the one-time broadcast).
+
+        A task exception does NOT crash this rank mid-collective (which
+        would strand the other ranks in the all-gather until timeout):
+        the failure travels through the gather and every rank raises the
+        same :class:`TaskFailedError` deterministically.
         """
         mine = self.shard_indices(len(tasks))
-        local = {i: task_fn(tasks[i]) for i in mine}
-        return self.gather_task_results(local, len(tasks))
+        local = {}
+        for i in mine:
+            if self.distributed:
+                try:
+                    local[i] = task_fn(tasks[i])
+                except Exception as e:  # noqa: BLE001 — re-raised post-gather
+                    import traceback
+
+                    local[i] = _TaskError(
+                        i, type(e).__name__, str(e),
+                        traceback.format_exc(),
+                    )
+            else:
+                local[i] = task_fn(tasks[i])
+        results = self.gather_task_results(local, len(tasks))
+        for r in results:
+            if isinstance(r, _TaskError):
+                raise TaskFailedError(
+                    f"task {r.task_id} failed on a worker rank with "
+                    f"{r.exc_type}: {r.msg}\n--- worker traceback ---\n"
+                    f"{r.tb}"
+                )
+        return results
 
     def gather_task_results(self, local, n_tasks):
         """all-gather {task_id: result} dicts → full ordered list."""

@@ -203,7 +203,31 @@ def _scenario_encoder(rank):
     return {"shape": [int(v) for v in T.shape]}
 
 
+def _scenario_task_failure(rank):
+    """A task that raises on ONE rank's shard must surface the SAME
+    error on EVERY rank after the gather (no hung collective)."""
+    from skdist_amd import Cluster
+    from skdist_amd.parallel.cluster import TaskFailedError
+
+    sc = Cluster()
+
+    def task_fn(i):
+        if i == 3:  # lands on exactly one rank's shard
+            raise ValueError("boom on task 3")
+        return i * i
+
+    try:
+        sc.run_tasks(task_fn, list(range(6)))
+        return {"raised": False}
+    except TaskFailedError as e:
+        # all ranks still in lock-step: a collective works after
+        sc.barrier()
+        return {"raised": True, "msg_has_type": "ValueError" in str(e),
+                "msg_has_id": "task 3" in str(e)}
+
+
 _SCENARIOS = {
+    "task_failure": _scenario_task_failure,
     "batched": _scenario_batched,
     "generic": _scenario_generic,
     "forest": _scenario_forest,
@@ -313,3 +337,11 @@ def test_spmd_encoder_gloo():
     outs = _run_spmd("encoder")
     assert outs[0]["shape"] == outs[1]["shape"]
     assert outs[0]["shape"][0] == 80
+
+
+@pytest.mark.timeout(900)
+def test_spmd_task_failure_gloo():
+    outs = _run_spmd("task_failure")
+    for r in (0, 1):
+        assert outs[r]["raised"]
+        assert outs[r]["msg_has_type"] and outs[r]["msg_has_id"]
