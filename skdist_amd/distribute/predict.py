@@ -83,13 +83,27 @@ class DistPredictor:
         per = (n + sc.world_size - 1) // sc.world_size
         lo = min(sc.rank * per, n)
         hi = min(lo + per, n)
-        mine = self._predict_local(X[lo:hi]) if hi > lo else None
+        try:
+            mine = self._predict_local(X[lo:hi]) if hi > lo else None
+            err = None
+        except Exception as e:  # noqa: BLE001 — re-raised post-gather on
+            import traceback  # every rank (no stranded collective)
+
+            mine = None
+            err = f"{type(e).__name__}: {e}\n{traceback.format_exc()}"
         # gather ordered shards from all ranks
         import torch.distributed as dist
 
+        from ..parallel.cluster import TaskFailedError
+
         boxes = [None] * sc.world_size
-        dist.all_gather_object(boxes, (sc.rank, mine))
-        boxes = [b for _, b in sorted(boxes) if b is not None]
+        dist.all_gather_object(boxes, (sc.rank, err, mine))
+        for _, e, _b in boxes:
+            if e is not None:
+                raise TaskFailedError(
+                    f"predict shard failed on a rank:\n{e}"
+                )
+        boxes = [b for _, _, b in sorted(boxes) if b is not None]
         return np.concatenate(boxes, axis=0)
 
     def _predict_local(self, X):
