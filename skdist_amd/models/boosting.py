@@ -19,6 +19,9 @@ binned dataset + level-synchronous tree builder as the forests
     tolerance rather than plain gradient-fitting;
   * ``subsample`` < 1 uses an exact 0/1 row mask through the builder's
     weight plane (stochastic gradient boosting).
+  * ``n_iter_no_change`` holds out ``validation_fraction`` rows (weight 0
+    through the same plane) and stops when their loss stalls for that
+    many rounds (sklearn's early-stopping semantics).
 
 Fitted state is host numpy only (HistTrees + priors): models pickle and
 predict like sklearn estimators and ride every meta-estimator's task
