@@ -56,10 +56,24 @@ class _BatchedLinearBase(BaseEstimator):
         return {"C"} if hasattr(self, "C") else {"alpha"}
 
     # ------------------------------------------------------------------ #
+    def _merged_sample_weight(self, y, sample_weight):
+        """Fold ``class_weight`` ('balanced' | dict) into per-row weights
+        (the solver's fused row-weight plane carries both)."""
+        cw = getattr(self, "class_weight", None)
+        if cw is None or y is None:
+            return sample_weight
+        from sklearn.utils.class_weight import compute_sample_weight
+
+        w = compute_sample_weight(cw, np.asarray(y))
+        if sample_weight is not None:
+            w = w * np.asarray(sample_weight, dtype=np.float64)
+        return w.astype(np.float32)
+
     def fit(self, X, y, sample_weight=None):
         t0 = time.perf_counter()
         sc = getattr(self, "sc", None)
         is_clf = isinstance(self, ClassifierMixin)
+        sample_weight = self._merged_sample_weight(y, sample_weight)
         ds = DeviceDataset(
             X, y,
             cluster=None,  # single fit: local device, no broadcast
@@ -161,6 +175,11 @@ class _BatchedLinearBase(BaseEstimator):
 
         is_clf = isinstance(self, ClassifierMixin)
         t0 = time.perf_counter()
+        # class_weight folds into the row-weight plane; computed from the
+        # FULL y (documented deviation: sklearn recomputes 'balanced' per
+        # training fold — fold class ratios match the full data for
+        # stratified/partition CV)
+        sample_weight = self._merged_sample_weight(y, sample_weight)
         ds = DeviceDataset(
             X, y,
             cluster=cluster, standardize=self.standardize,
@@ -455,6 +474,10 @@ class _BatchedLinearBase(BaseEstimator):
         with estimators[i] a fitted BINARY copy of self (classes_=[0,1],
         positive class = the OvR class / the pair's second class).
         """
+        if getattr(self, "class_weight", None) is not None:
+            raise FallbackToGeneric(
+                "class_weight is per-binary-problem: generic path"
+            )
         t0 = time.perf_counter()
         ds = DeviceDataset(
             X, y,
@@ -558,8 +581,9 @@ class LogisticRegression(ClassifierMixin, _BatchedLinearBase):
 
     def __init__(self, C=1.0, lr=0.5, epochs=20, batch_size=8192,
                  momentum=0.9, lr_decay=0.0, standardize=True,
-                 random_state=None, sc=None):
+                 class_weight=None, random_state=None, sc=None):
         self.C = C
+        self.class_weight = class_weight
         self.lr_decay = lr_decay
         self.lr = lr
         self.epochs = epochs
@@ -595,8 +619,9 @@ class LinearSVC(ClassifierMixin, _BatchedLinearBase):
 
     def __init__(self, C=1.0, lr=0.5, epochs=20, batch_size=8192,
                  momentum=0.9, lr_decay=0.0, standardize=True,
-                 random_state=None, sc=None):
+                 class_weight=None, random_state=None, sc=None):
         self.C = C
+        self.class_weight = class_weight
         self.lr_decay = lr_decay
         self.lr = lr
         self.epochs = epochs
