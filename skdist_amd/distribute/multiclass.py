@@ -116,7 +116,8 @@ class DistOneVsRestClassifier(ClassifierMixin, BaseEstimator):
 
     def __init__(self, estimator, sc=None, norm=None, partitions="auto",
                  max_negatives=None, random_state=None, method="ratio",
-                 n_splits=1, mlb_override=False, verbose=False):
+                 n_splits=1, mlb_override=False, verbose=False,
+                 n_jobs=None):
         self.estimator = estimator
         self.sc = sc
         self.norm = norm
@@ -127,6 +128,7 @@ class DistOneVsRestClassifier(ClassifierMixin, BaseEstimator):
         self.n_splits = n_splits  # API parity; HBM needs no chunked bcast
         self.mlb_override = mlb_override
         self.verbose = verbose
+        self.n_jobs = n_jobs
 
     # ------------------------------------------------------------------ #
     def fit(self, X, y, **fit_params):
@@ -188,7 +190,7 @@ class DistOneVsRestClassifier(ClassifierMixin, BaseEstimator):
 
         tasks = list(enumerate(columns))
         if sc is None:
-            results = run_local_tasks(task_fn, tasks)
+            results = run_local_tasks(task_fn, tasks, n_jobs=self.n_jobs)
         else:
             results = sc.run_tasks(task_fn, tasks)
         results.sort(key=lambda t: t[0])
@@ -330,11 +332,13 @@ class DistOneVsOneClassifier(ClassifierMixin, BaseEstimator):
     """One-vs-one with distributed per-pair fits
     (reference multiclass.py:365-475)."""
 
-    def __init__(self, estimator, sc=None, partitions="auto", verbose=False):
+    def __init__(self, estimator, sc=None, partitions="auto", verbose=False,
+                 n_jobs=None):
         self.estimator = estimator
         self.sc = sc
         self.partitions = partitions
         self.verbose = verbose
+        self.n_jobs = n_jobs
 
     def fit(self, X, y, **fit_params):
         _check_estimator(self, verbose=self.verbose)
@@ -380,7 +384,7 @@ class DistOneVsOneClassifier(ClassifierMixin, BaseEstimator):
 
         tasks = list(enumerate(pairs))
         if sc is None:
-            results = run_local_tasks(task_fn, tasks)
+            results = run_local_tasks(task_fn, tasks, n_jobs=self.n_jobs)
         else:
             results = sc.run_tasks(task_fn, tasks)
         results.sort(key=lambda t: t[0])
