@@ -26,6 +26,7 @@ MODULES = [
     "skdist_amd.models.linear",
     "skdist_amd.models._sgd",
     "skdist_amd.models.forest",
+    "skdist_amd.models.boosting",
     "skdist_amd.ops",
     "skdist_amd.ops.build",
 ]
@@ -63,7 +64,18 @@ def test_public_api_surface():
         DistMultiModelSearch,
         DistRandomizedSearchCV,
     )
-    from skdist_amd.models import LinearSVC, LogisticRegression, Ridge  # noqa: F401
+    from skdist_amd.models import (  # noqa: F401
+        HistGradientBoostingClassifier,
+        HistGradientBoostingRegressor,
+        LinearSVC,
+        LogisticRegression,
+        Ridge,
+    )
+    from skdist_amd.distribute.ensemble import (  # noqa: F401
+        get_oof,
+        get_single_oof,
+    )
+    from skdist_amd.parallel.cluster import TaskFailedError  # noqa: F401
     from skdist_amd.postprocessing import SimpleVoter  # noqa: F401
     from skdist_amd.preprocessing import (  # noqa: F401
         DenseTransformer,

@@ -87,3 +87,30 @@ def test_multimodel_batched_path_matches_generic():
         mm_b.cv_results_["mean_test_score"],
         mm_g.cv_results_["mean_test_score"], atol=0.05)
     assert mm_b.best_score_ > 0.8
+
+
+def test_multimodel_with_boosted_family():
+    """Heterogeneous pool mixing the native linear and boosted families
+    (reference's pool mixed sklearn + xgboost, search.py:717-908)."""
+    from skdist_amd.models import (
+        HistGradientBoostingClassifier,
+        LogisticRegression,
+    )
+
+    rng = np.random.default_rng(0)
+    X = rng.standard_normal((600, 6)).astype(np.float32)
+    y = (np.sin(X[:, 0]) + X[:, 1] > 0).astype(int)
+    mm = DistMultiModelSearch(
+        [
+            ("lr", LogisticRegression(epochs=8, random_state=0),
+             {"C": [0.1, 1.0]}),
+            ("gbt", HistGradientBoostingClassifier(
+                n_estimators=20, random_state=0),
+             {"max_depth": [2, 3]}),
+        ],
+        n=2, cv=3, random_state=0,
+    )
+    mm.fit(X, y)
+    assert mm.best_score_ > 0.85
+    assert mm.best_model_name_ in ("lr", "gbt")
+    assert mm.predict(X[:4]).shape == (4,)
