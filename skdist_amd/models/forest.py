@@ -1165,8 +1165,10 @@ def flat_forest_for(model, device):
     except Exception:
         pass
     trees = getattr(model, "estimators_", None)
-    if not trees:
-        return None
+    if trees is None or len(trees) == 0 or getattr(trees, "ndim", 1) != 1:
+        return None  # (a boosted model's [n_stages, K] array is not a
+        # flat forest — those models expose _device_predict_fn instead)
+    trees = list(trees)
     if all(isinstance(t, HistTree) for t in trees):
         return FlatForest(trees, device)
     if all(hasattr(t, "tree_") for t in trees):
