@@ -226,7 +226,43 @@ def _scenario_task_failure(rank):
                 "msg_has_id": "task 3" in str(e)}
 
 
+def _scenario_multimodel(rank):
+    """Heterogeneous model pool at world 2: the batched per-model hook
+    shards candidates, generic families shard tasks; every rank
+    assembles identical results."""
+    from sklearn.tree import DecisionTreeClassifier
+
+    from skdist_amd import Cluster
+    from skdist_amd.distribute.search import DistMultiModelSearch
+    from skdist_amd.models import LogisticRegression
+
+    sc = Cluster()
+    X = y = None
+    if rank == 0:
+        rng = np.random.default_rng(5)
+        X = rng.standard_normal((700, 6)).astype(np.float32)
+        y = ((X[:, 0] + X[:, 1] > 0)).astype(np.int64)
+    mm = DistMultiModelSearch(
+        [
+            ("lr", LogisticRegression(epochs=8, random_state=0),
+             {"C": [0.1, 1.0, 10.0]}),
+            ("tree", DecisionTreeClassifier(random_state=0),
+             {"max_depth": [2, 4]}),
+        ],
+        n=2, cv=3, random_state=0, sc=sc,
+    )
+    mm.fit(X, y)
+    return {
+        "best": mm.best_model_name_,
+        "scores": [float(v) for v in mm.cv_results_["mean_test_score"]],
+        "acc": float(
+            (mm.predict(sc.sync_host_data(X))
+             == sc.sync_host_data(y)).mean()),
+    }
+
+
 _SCENARIOS = {
+    "multimodel": _scenario_multimodel,
     "task_failure": _scenario_task_failure,
     "batched": _scenario_batched,
     "generic": _scenario_generic,
@@ -345,3 +381,11 @@ def test_spmd_task_failure_gloo():
     for r in (0, 1):
         assert outs[r]["raised"]
         assert outs[r]["msg_has_type"] and outs[r]["msg_has_id"]
+
+
+@pytest.mark.timeout(900)
+def test_spmd_multimodel_gloo():
+    outs = _run_spmd("multimodel")
+    assert np.allclose(outs[0]["scores"], outs[1]["scores"])
+    assert outs[0]["best"] == outs[1]["best"]
+    assert outs[0]["acc"] > 0.9
