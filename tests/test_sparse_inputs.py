@@ -113,3 +113,38 @@ def test_text_pipeline_sparse_end_to_end():
     gs.fit(Xs, y)
     assert gs.best_score_ > 0.95
     assert (gs.predict(Xs) == y).mean() > 0.95
+
+
+def test_encoderizer_to_native_ovr_text_flow():
+    """The reference's flagship text workflow: Encoderizer sparse text
+    features -> native OvR (densified into the device dataset), with a
+    pickled (encoder, model) pair predicting end-to-end."""
+    import pandas as pd
+
+    from skdist_amd.distribute.encoder import Encoderizer
+    from skdist_amd.distribute.multiclass import DistOneVsRestClassifier
+    from skdist_amd.models import LinearSVC
+
+    rng = np.random.default_rng(0)
+    topics = ["sports ball game team win",
+              "market stock trade price fund",
+              "code python bug compile test"]
+    docs, labels = [], []
+    for i in range(600):
+        k = i % 3
+        docs.append(" ".join(
+            rng.choice(topics[k].split(), size=6)) + f" common{i % 5}")
+        labels.append(["sport", "finance", "tech"][k])
+    df = pd.DataFrame({"text": docs, "num": rng.standard_normal(600)})
+    y = np.array(labels)
+
+    enc = Encoderizer(size="medium")
+    T = enc.fit_transform(df)
+    assert sp.issparse(T)
+    ovr = DistOneVsRestClassifier(
+        LinearSVC(epochs=12, random_state=0)
+    ).fit(T, y)
+    assert (ovr.predict(enc.transform(df)) == y).mean() > 0.98
+    blob = pickle.loads(pickle.dumps({"enc": enc, "ovr": ovr}))
+    out = blob["ovr"].predict(blob["enc"].transform(df.iloc[:6]))
+    assert set(out) <= {"sport", "finance", "tech"}
