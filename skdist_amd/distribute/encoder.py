@@ -152,6 +152,20 @@ class Encoderizer(TransformerMixin, BaseEstimator):
 
         registry = _default_encoders[self.size]
         if self.config is not None:
+            bad_cols = [c for c in self.config if c not in X.columns]
+            if bad_cols:
+                raise ValueError(
+                    f"config names columns not in the input: {bad_cols} "
+                    f"(available: {list(X.columns)})"
+                )
+            bad_kinds = {
+                k for k in self.config.values() if k not in registry
+            }
+            if bad_kinds:
+                raise ValueError(
+                    f"unknown encoder kind(s) {sorted(bad_kinds)}; "
+                    f"choose from {sorted(registry)}"
+                )
             groups = [registry[kind](col) for col, kind in
                       self.config.items()]
         else:

@@ -143,3 +143,11 @@ def test_list_of_record_dicts():
     T = enc.fit_transform(recs)
     assert T.shape[0] == 40
     assert enc.transform(recs[:5]).shape == (5, T.shape[1])
+
+
+def test_config_validation_errors():
+    df = pd.DataFrame({"a": [1.0, 2.0], "b": ["x", "y"]})
+    with pytest.raises(ValueError, match="not in the input"):
+        Encoderizer(size="small", config={"zzz": "numeric"}).fit(df)
+    with pytest.raises(ValueError, match="unknown encoder kind"):
+        Encoderizer(size="small", config={"a": "bogus"}).fit(df)
