@@ -400,6 +400,11 @@ class DistOneVsOneClassifier(ClassifierMixin, BaseEstimator):
             fn = self._device_predict_fn("predict", "cuda")
             if fn is not None:
                 return fn(X)
+        return self.classes_[self.decision_function(X).argmax(axis=1)]
+
+    def decision_function(self, X):
+        """Votes + bounded confidence tie-break per class — sklearn's
+        OvO ``_ovr_decision_function`` semantics (argmax == predict)."""
         k = len(self.classes_)
         n = _n_rows(X)
         votes = np.zeros((n, k))
@@ -414,21 +419,7 @@ class DistOneVsOneClassifier(ClassifierMixin, BaseEstimator):
             votes[~pred_j, i] += 1
             conf[:, j] += d
             conf[:, i] -= d
-        # tie-break votes with summed confidences (sklearn semantics)
-        score = votes + conf / (3 * (np.abs(conf) + 1))
-        return self.classes_[score.argmax(axis=1)]
-
-    def decision_function(self, X):
-        k = len(self.classes_)
-        conf = np.zeros((_n_rows(X), k))
-        for est, (i, j) in zip(self.estimators_, self.pairs_):
-            if hasattr(est, "decision_function"):
-                d = np.asarray(est.decision_function(X), dtype=float)
-            else:
-                d = est.predict_proba(X)[:, -1] - 0.5
-            conf[:, j] += d
-            conf[:, i] -= d
-        return conf
+        return votes + conf / (3 * (np.abs(conf) + 1))
 
     # batched device inference (DistPredictor hook): all k(k-1)/2 pair
     # models score as ONE GPU GEMM, votes/confidences fused on device

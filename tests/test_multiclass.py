@@ -174,3 +174,19 @@ def test_negatives_mask_modes():
     assert m.sum() == n_pos + 100
     # all positives always kept
     assert m[y == 1].all()
+
+
+def test_ovo_decision_function_matches_sklearn():
+    """OvO decision_function == sklearn's _ovr_decision_function
+    (votes + bounded confidence tie-break), not raw confidence sums."""
+    from sklearn.multiclass import OneVsOneClassifier
+
+    rng = np.random.default_rng(0)
+    X = rng.standard_normal((400, 8)).astype(np.float32)
+    y = np.array(["aa", "bb", "cc", "dd"])[X[:, :4].argmax(axis=1)]
+    ours = DistOneVsOneClassifier(SkLogReg(solver="liblinear")).fit(X, y)
+    ref = OneVsOneClassifier(SkLogReg(solver="liblinear")).fit(X, y)
+    np.testing.assert_allclose(
+        ours.decision_function(X), ref.decision_function(X), atol=1e-12
+    )
+    assert (ours.predict(X) == ref.predict(X)).all()

@@ -8,7 +8,6 @@ must never change any column's result.
 """
 
 import numpy as np
-import pytest
 import torch
 from hypothesis import given, settings
 from hypothesis import strategies as st
