@@ -261,7 +261,13 @@ class DistOneVsRestClassifier(ClassifierMixin, BaseEstimator):
         return probs
 
     def decision_function(self, X):
-        scores = self._scores(X)
+        """Per-class raw scores from each binary estimator's
+        decision_function (sklearn OvR semantics — NOT probabilities;
+        predict/predict_proba use the proba-preferring ``_scores``)."""
+        scores = np.column_stack([
+            np.asarray(est.decision_function(X), dtype=float)
+            for est in self.estimators_
+        ])
         if self._binary_single_column():
             return scores.ravel()
         return scores

@@ -190,3 +190,25 @@ def test_ovo_decision_function_matches_sklearn():
         ours.decision_function(X), ref.decision_function(X), atol=1e-12
     )
     assert (ours.predict(X) == ref.predict(X)).all()
+
+
+def test_ovr_decision_function_matches_sklearn():
+    """OvR decision_function = per-class RAW estimator scores (sklearn
+    semantics), while predict_proba stays the normalized-proba matrix."""
+    from sklearn.multiclass import OneVsRestClassifier
+
+    rng = np.random.default_rng(0)
+    X = rng.standard_normal((300, 8)).astype(np.float32)
+    y = X[:, :3].argmax(axis=1)
+    ours = DistOneVsRestClassifier(SkLogReg(solver="liblinear")).fit(X, y)
+    ref = OneVsRestClassifier(SkLogReg(solver="liblinear")).fit(X, y)
+    np.testing.assert_allclose(
+        ours.decision_function(X), ref.decision_function(X), atol=1e-12
+    )
+    yb = (X[:, 0] > 0).astype(int)
+    ob = DistOneVsRestClassifier(SkLogReg(solver="liblinear")).fit(X, yb)
+    rb = OneVsRestClassifier(SkLogReg(solver="liblinear")).fit(X, yb)
+    assert ob.decision_function(X).shape == (300,)
+    np.testing.assert_allclose(
+        ob.decision_function(X), rb.decision_function(X), atol=1e-12
+    )
