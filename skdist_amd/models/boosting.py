@@ -188,6 +188,22 @@ class _BaseHistGB(BaseEstimator):
                 )
         return out
 
+    def _staged_raw(self, X):
+        """Yield raw scores after each boosting round (sklearn's
+        staged_* protocol; one traversal per stage, accumulated)."""
+        import scipy.sparse as sp
+
+        if sp.issparse(X):
+            X = X.toarray()
+        X = np.asarray(X, dtype=np.float32)
+        out = np.tile(self._base_raw(), (X.shape[0], 1))
+        for round_trees in self.stages_:
+            for k, tree in enumerate(round_trees):
+                out[:, k] += (
+                    self.learning_rate * tree.value[_leaf_rows(tree, X), 0]
+                )
+            yield out.copy()
+
     @property
     def estimators_(self):
         """[n_stages, K] object array of HistTrees (sklearn-shaped)."""
@@ -284,6 +300,10 @@ class HistGradientBoostingRegressor(RegressorMixin, _BaseHistGB):
     def predict(self, X):
         return self._raw_scores(X)[:, 0]
 
+    def staged_predict(self, X):
+        for r in self._staged_raw(X):
+            yield r[:, 0]
+
 
 class HistGradientBoostingClassifier(ClassifierMixin, _BaseHistGB):
     """Binomial / multinomial deviance gradient boosting on binned
@@ -355,3 +375,25 @@ class HistGradientBoostingClassifier(ClassifierMixin, _BaseHistGB):
         if r.shape[1] == 1:
             return self.classes_[(r[:, 0] > 0).astype(np.int64)]
         return self.classes_[r.argmax(axis=1)]
+
+    def _proba_from_raw(self, r):
+        if r.shape[1] == 1:
+            p = 1.0 / (1.0 + np.exp(-r[:, 0]))
+            return np.column_stack([1.0 - p, p])
+        e = np.exp(r - r.max(axis=1, keepdims=True))
+        return e / e.sum(axis=1, keepdims=True)
+
+    def staged_predict_proba(self, X):
+        for r in self._staged_raw(X):
+            yield self._proba_from_raw(r)
+
+    def staged_predict(self, X):
+        for r in self._staged_raw(X):
+            if r.shape[1] == 1:
+                yield self.classes_[(r[:, 0] > 0).astype(np.int64)]
+            else:
+                yield self.classes_[r.argmax(axis=1)]
+
+    def staged_decision_function(self, X):
+        for r in self._staged_raw(X):
+            yield r[:, 0] if r.shape[1] == 1 else r

@@ -147,3 +147,28 @@ def test_early_stopping(data):
         n_estimators=300, n_iter_no_change=5, random_state=0
     ).fit(X, X[:, 0] * 2.0)
     assert r.n_estimators_ < 300
+
+
+def test_staged_predictions(data):
+    """staged_* iterators (sklearn protocol): stage i equals a model
+    trained with i+1 rounds; the last stage equals predict."""
+    X, t = data
+    y = (t > np.median(t)).astype(int)
+    m = HistGradientBoostingClassifier(
+        n_estimators=8, random_state=0
+    ).fit(X, y)
+    stages = list(m.staged_predict_proba(X))
+    assert len(stages) == 8
+    np.testing.assert_allclose(stages[-1], m.predict_proba(X))
+    m3 = HistGradientBoostingClassifier(
+        n_estimators=3, random_state=0
+    ).fit(X, y)
+    np.testing.assert_allclose(stages[2], m3.predict_proba(X), atol=1e-6)
+    preds = list(m.staged_predict(X))
+    assert (preds[-1] == m.predict(X)).all()
+    r = HistGradientBoostingRegressor(n_estimators=5, random_state=0).fit(
+        X, t
+    )
+    rs = list(r.staged_predict(X))
+    assert len(rs) == 5
+    np.testing.assert_allclose(rs[-1], r.predict(X))
