@@ -360,12 +360,7 @@ class HistGradientBoostingClassifier(ClassifierMixin, _BaseHistGB):
         return r[:, 0] if r.shape[1] == 1 else r
 
     def predict_proba(self, X):
-        r = self._raw_scores(X)
-        if r.shape[1] == 1:
-            p = 1.0 / (1.0 + np.exp(-r[:, 0]))
-            return np.column_stack([1.0 - p, p])
-        e = np.exp(r - r.max(axis=1, keepdims=True))
-        return e / e.sum(axis=1, keepdims=True)
+        return self._proba_from_raw(self._raw_scores(X))
 
     def predict_log_proba(self, X):
         return np.log(np.clip(self.predict_proba(X), 1e-300, None))
