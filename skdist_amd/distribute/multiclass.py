@@ -233,7 +233,13 @@ class DistOneVsRestClassifier(ClassifierMixin, BaseEstimator):
                 return out
         scores = self._scores(X)
         if self.mlb_ is not None or getattr(self, "multilabel_", False):
-            ind = (scores > 0.5).astype(int)
+            # _scores mixes probability columns (threshold 0.5) and raw
+            # decision columns (threshold 0) depending on the estimator
+            thr = np.array([
+                0.5 if hasattr(est, "predict_proba") else 0.0
+                for est in self.estimators_
+            ])
+            ind = (scores > thr[None, :]).astype(int)
             if self.mlb_ is not None:
                 return self.mlb_.inverse_transform(ind)
             return ind

@@ -212,3 +212,24 @@ def test_ovr_decision_function_matches_sklearn():
     np.testing.assert_allclose(
         ob.decision_function(X), rb.decision_function(X), atol=1e-12
     )
+
+
+def test_multilabel_thresholds_match_sklearn():
+    """Multilabel indicator predictions: probability columns threshold
+    at 0.5, raw decision columns at 0 (regression: a flat 0.5 threshold
+    under-predicted positives for no-proba estimators)."""
+    from sklearn.multiclass import OneVsRestClassifier
+    from sklearn.svm import LinearSVC as SkSVC
+
+    rng = np.random.default_rng(0)
+    X = rng.standard_normal((300, 8)).astype(np.float32)
+    Y = np.zeros((300, 3), dtype=int)
+    Y[:, 0] = X[:, 0] > 0
+    Y[:, 1] = X[:, 1] > 0.5
+    Y[:, 2] = (X[:, 0] + X[:, 1]) > 0
+    for est_f in (SkSVC, lambda: SkLogReg(solver="liblinear")):
+        ours = DistOneVsRestClassifier(est_f()).fit(X, Y)
+        ref = OneVsRestClassifier(est_f()).fit(X, Y)
+        np.testing.assert_array_equal(
+            np.asarray(ours.predict(X)), ref.predict(X)
+        )
