@@ -128,3 +128,41 @@ def test_feature_mask_column_independence():
     W_m2 = solve(m2[:, None])
     np.testing.assert_allclose(W_joint[:, 1:], W_m2, rtol=1e-4, atol=1e-6)
     assert (W_joint[[2, 4], 1] == 0).all()  # masked rows pinned to zero
+
+
+@settings(max_examples=15, deadline=None)
+@given(
+    seed=st.integers(0, 10_000),
+    cv=st.integers(2, 4),
+    k=st.integers(2, 3),
+    grid_n=st.integers(1, 4),
+)
+def test_generic_search_equals_sklearn_gridsearch(seed, cv, k, grid_n):
+    """The generic fan-out is sklearn's GridSearchCV, exactly: same
+    mean scores, same best candidate, on random tiny problems."""
+    import warnings
+
+    from sklearn.linear_model import LogisticRegression as SkLR
+    from sklearn.model_selection import GridSearchCV
+
+    from skdist_amd.distribute.search import DistGridSearchCV
+
+    rng = np.random.default_rng(seed)
+    n = int(rng.integers(60, 200))
+    f = int(rng.integers(2, 6))
+    X = rng.standard_normal((n, f))
+    y = rng.integers(0, k, n)
+    y[: k] = np.arange(k)  # every class present
+    grid = {"C": sorted(rng.uniform(0.01, 10, grid_n).tolist())}
+    with warnings.catch_warnings():
+        warnings.simplefilter("ignore")
+        ours = DistGridSearchCV(
+            SkLR(solver="liblinear"), grid, cv=cv
+        ).fit(X, y)
+        ref = GridSearchCV(SkLR(solver="liblinear"), grid, cv=cv).fit(X, y)
+    np.testing.assert_allclose(
+        ours.cv_results_["mean_test_score"],
+        ref.cv_results_["mean_test_score"],
+    )
+    assert ours.best_index_ == ref.best_index_
+    np.testing.assert_allclose(ours.best_score_, ref.best_score_)
