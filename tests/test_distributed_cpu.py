@@ -261,7 +261,32 @@ def _scenario_multimodel(rank):
     }
 
 
+def _scenario_ovo(rank):
+    from skdist_amd import Cluster
+    from skdist_amd.distribute.multiclass import DistOneVsOneClassifier
+    from skdist_amd.models import LogisticRegression
+
+    sc = Cluster()
+    X = y = None
+    if rank == 0:
+        rng = np.random.default_rng(6)
+        X = rng.standard_normal((800, 8)).astype(np.float32)
+        W = rng.standard_normal((4, 8))
+        y = (X @ W.T).argmax(axis=1)
+    ovo = DistOneVsOneClassifier(
+        LogisticRegression(epochs=10, random_state=0), sc=sc)
+    ovo.fit(X, y)
+    Xh = sc.sync_host_data(X)
+    df = ovo.decision_function(Xh)
+    return {
+        "n_pairs": len(ovo.estimators_),
+        "acc": float((ovo.predict(Xh) == sc.sync_host_data(y)).mean()),
+        "df0": [float(v) for v in df[0]],
+    }
+
+
 _SCENARIOS = {
+    "ovo": _scenario_ovo,
     "multimodel": _scenario_multimodel,
     "task_failure": _scenario_task_failure,
     "batched": _scenario_batched,
@@ -389,3 +414,11 @@ def test_spmd_multimodel_gloo():
     assert np.allclose(outs[0]["scores"], outs[1]["scores"])
     assert outs[0]["best"] == outs[1]["best"]
     assert outs[0]["acc"] > 0.9
+
+
+@pytest.mark.timeout(900)
+def test_spmd_ovo_gloo():
+    outs = _run_spmd("ovo")
+    assert outs[0]["n_pairs"] == 6
+    assert outs[0]["acc"] > 0.9
+    assert np.allclose(outs[0]["df0"], outs[1]["df0"])
