@@ -23,7 +23,7 @@ search = DistRandomizedSearchCV(
         "learning_rate": [0.05, 0.1, 0.3],
         "subsample": [0.7, 1.0],
     },
-    n_iter=6, cv=5, scoring="roc_auc", random_state=0, sc=None,
+    n_iter=4, cv=3, scoring="roc_auc", random_state=0, sc=None,
 )
 search.fit(X.astype(np.float32), y)
 print("best params:", search.best_params_)
