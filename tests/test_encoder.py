@@ -151,3 +151,18 @@ def test_config_validation_errors():
         Encoderizer(size="small", config={"zzz": "numeric"}).fit(df)
     with pytest.raises(ValueError, match="unknown encoder kind"):
         Encoderizer(size="small", config={"a": "bogus"}).fit(df)
+
+
+def test_defaults_registry_factories():
+    """Every tier's factory returns [(name, pipeline)] steps (reference
+    test_defaults.py:18-57) and the identity tokenizer is identity."""
+    from skdist_amd.distribute import _defaults
+
+    assert _defaults.tokenizer(5) == 5
+    for size in ("small", "medium", "large"):
+        for kind, factory in _defaults._default_encoders[size].items():
+            steps = factory("c")
+            assert isinstance(steps, list) and steps
+            for name, pipe in steps:
+                assert "c" in name
+                assert hasattr(pipe, "fit") and hasattr(pipe, "transform")
