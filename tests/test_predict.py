@@ -75,3 +75,29 @@ def test_prediction_fn_text():
     fn = get_prediction_fn(model, method="predict", feature_type="text")
     preds = fn(docs[:4])
     assert list(preds) == [1, 0, 1, 0]
+
+
+def test_predictor_decision_function_and_gbt_fn():
+    """DistPredictor(method='decision_function') and get_prediction_fn
+    over the native boosted family (pandas feature_type)."""
+    from sklearn.linear_model import LogisticRegression as SkLR
+
+    from skdist_amd.models import HistGradientBoostingClassifier
+
+    rng = np.random.default_rng(0)
+    X = rng.standard_normal((1000, 6)).astype(np.float32)
+    y = (X[:, 0] > 0).astype(int)
+    m = SkLR(solver="liblinear").fit(X, y)
+    p = DistPredictor(m, method="decision_function", chunk_rows=128)
+    np.testing.assert_allclose(p(X), m.decision_function(X))
+
+    gbt = HistGradientBoostingClassifier(
+        n_estimators=10, random_state=0
+    ).fit(X, y)
+    fn = get_prediction_fn(
+        gbt, method="predict_proba", feature_type="pandas",
+        names=[f"f{i}" for i in range(6)],
+    )
+    np.testing.assert_allclose(
+        fn(*[X[:, j] for j in range(6)]), gbt.predict_proba(X)
+    )
