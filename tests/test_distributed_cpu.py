@@ -422,3 +422,51 @@ def test_spmd_ovo_gloo():
     assert outs[0]["n_pairs"] == 6
     assert outs[0]["acc"] > 0.9
     assert np.allclose(outs[0]["df0"], outs[1]["df0"])
+
+
+@pytest.mark.timeout(300)
+def test_pickle_loads_in_fresh_interpreter(tmp_path):
+    """A model fitted WITH a Cluster must unpickle and predict in a
+    clean process with no torch.distributed context at all (the
+    serve-side contract: fit on the GPU box, score anywhere)."""
+    import subprocess
+    import sys
+
+    import numpy as np
+
+    from skdist_amd import Cluster
+    from skdist_amd.distribute.search import DistGridSearchCV
+    from skdist_amd.models import LogisticRegression
+
+    rng = np.random.default_rng(0)
+    X = rng.standard_normal((300, 6)).astype(np.float32)
+    y = (X[:, 0] > 0).astype(np.int64)
+    gs = DistGridSearchCV(
+        LogisticRegression(epochs=8, random_state=0),
+        {"C": [0.5, 2.0]}, cv=3, sc=Cluster(),
+    ).fit(X, y)
+    blob = tmp_path / "model.pkl"
+    xfile = tmp_path / "X.npy"
+    np.save(xfile, X)
+    import pickle
+
+    blob.write_bytes(pickle.dumps(gs))
+    code = (
+        "import pickle, sys, numpy as np\n"
+        f"m = pickle.loads(open(r'{blob}', 'rb').read())\n"
+        f"X = np.load(r'{xfile}')\n"
+        "p = m.predict(X)\n"
+        "print('OK', (p == (X[:, 0] > 0)).mean())\n"
+    )
+    import os
+
+    repo_root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    out = subprocess.run(
+        [sys.executable, "-c", code], capture_output=True, text=True,
+        timeout=240, cwd="/",  # away from the repo: only PYTHONPATH imports
+        env={**os.environ, "PYTHONPATH": repo_root},
+    )
+    assert out.returncode == 0, out.stderr[-1000:]
+    assert "OK" in out.stdout
+    acc = float(out.stdout.split()[-1])
+    assert acc > 0.9
