@@ -23,6 +23,12 @@ binned dataset + level-synchronous tree builder as the forests
     through the same plane) and stops when their loss stalls for that
     many rounds (sklearn's early-stopping semantics).
 
+The CPU path runs the builder's eager torch mirror — correct and
+usable (50 trees on 100k x 20 in ~8 s) but, like the rest of the CPU
+tier, a fallback: the production path is the HIP builder on the GPU
+(sklearn's dedicated C `HistGradientBoosting*` remains the right tool
+for CPU-only deployments).
+
 Fitted state is host numpy only (HistTrees + priors): models pickle and
 predict like sklearn estimators and ride every meta-estimator's task
 fan-out (DistGridSearchCV and friends) exactly as the reference's
