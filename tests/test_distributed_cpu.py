@@ -285,7 +285,32 @@ def _scenario_ovo(rank):
     }
 
 
+def _scenario_ridge(rank):
+    """Batched REGRESSION solve at world 2 (task='reg' label path +
+    r2 scoring through the batched device metrics)."""
+    from skdist_amd import Cluster
+    from skdist_amd.distribute.search import DistGridSearchCV
+    from skdist_amd.models import Ridge
+
+    sc = Cluster()
+    X = y = None
+    if rank == 0:
+        rng = np.random.default_rng(7)
+        X = rng.standard_normal((2000, 8)).astype(np.float32)
+        y = (X @ rng.standard_normal(8)
+             + 0.1 * rng.standard_normal(2000)).astype(np.float64)
+    gs = DistGridSearchCV(
+        Ridge(epochs=12, random_state=0),
+        {"alpha": [0.1, 1.0, 10.0]}, cv=3, scoring="r2", sc=sc)
+    gs.fit(X, y)
+    return {
+        "best": gs.best_score_,
+        "scores": [float(v) for v in gs.cv_results_["mean_test_score"]],
+    }
+
+
 _SCENARIOS = {
+    "ridge": _scenario_ridge,
     "ovo": _scenario_ovo,
     "multimodel": _scenario_multimodel,
     "task_failure": _scenario_task_failure,
@@ -470,3 +495,10 @@ def test_pickle_loads_in_fresh_interpreter(tmp_path):
     assert "OK" in out.stdout
     acc = float(out.stdout.split()[-1])
     assert acc > 0.9
+
+
+@pytest.mark.timeout(900)
+def test_spmd_ridge_gloo():
+    outs = _run_spmd("ridge")
+    assert outs[0]["best"] > 0.95
+    assert np.allclose(outs[0]["scores"], outs[1]["scores"])
