@@ -59,8 +59,8 @@ def _scenario_batched(rank):
         X, y = load_breast_cancer(return_X_y=True)
     gs = DistGridSearchCV(
         LogisticRegression(epochs=10, random_state=0),
-        {"C": [0.01, 0.1, 1.0, 10.0]},
-        cv=3, scoring="roc_auc", sc=sc,
+        {"C": [0.01, 0.1, 1.0, 10.0, 100.0]},  # 5 candidates over 2
+        cv=3, scoring="roc_auc", sc=sc,        # ranks: uneven shard
     )
     gs.fit(X, y)
     blob = pickle.dumps(gs)  # sc stripped, must pickle
@@ -368,7 +368,7 @@ def test_spmd_batched_gloo():
     X, y = load_breast_cancer(return_X_y=True)
     gs = DistGridSearchCV(
         LogisticRegression(epochs=10, random_state=0),
-        {"C": [0.01, 0.1, 1.0, 10.0]},
+        {"C": [0.01, 0.1, 1.0, 10.0, 100.0]},
         cv=3, scoring="roc_auc", sc=Cluster(),
     )
     gs.fit(X, y)
