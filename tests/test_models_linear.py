@@ -265,3 +265,27 @@ def test_class_weight_through_batched_search():
     )
     gs.fit(X, y)
     assert gs.best_score_ > 0.9
+
+
+def test_default_epochs_handle_tiny_overparameterized_data():
+    """The 20-epoch default exists for small data: on a 50x40
+    noise-dominated task the native solver at defaults must be at
+    sklearn-liblinear CV parity (bench configs lower epochs only
+    because 1M-row tasks converge in fewer passes)."""
+    from sklearn.linear_model import LogisticRegression as SkLR
+    from sklearn.model_selection import GridSearchCV
+
+    from skdist_amd import Cluster
+    from skdist_amd.distribute.search import DistGridSearchCV
+
+    rng = np.random.default_rng(1)
+    X = rng.standard_normal((50, 40)).astype(np.float32)
+    y = (X[:, 0] > 0).astype(int)
+    ref = GridSearchCV(
+        SkLR(solver="liblinear"), {"C": [0.5, 2.0]}, cv=3
+    ).fit(X, y)
+    ours = DistGridSearchCV(
+        LogisticRegression(random_state=0),  # default epochs=20
+        {"C": [0.5, 2.0]}, cv=3, sc=Cluster(),
+    ).fit(X, y)
+    assert ours.best_score_ > ref.best_score_ - 0.05
