@@ -98,11 +98,30 @@ def get_oof(clf, X, y, n_splits=5, sc=None):
 
     y = np.asarray(y)
     folds = list(KFold(n_splits=n_splits).split(X))
-    oof = np.zeros((y.shape[0], len(np.unique(y))))
+    classes = np.unique(y)
+    oof = np.zeros((y.shape[0], len(classes)))
 
     def task_fn(task):
+        # like get_single_oof, but fold probabilities are aligned to the
+        # GLOBAL class set via est.classes_: an unshuffled KFold on
+        # label-sorted data can miss a class in a training fold, which
+        # would otherwise shape-error on assignment (latent in the
+        # reference too, ensemble.py:130-151)
+        from .base import _clone
+        from .utils import _safe_split
+
         train_idx, test_idx = task
-        return get_single_oof(clf, X, y, train_idx, test_idx)
+        est = _clone(clf)
+        if hasattr(est, "sc"):
+            est.sc = None
+        X_tr, y_tr = _safe_split(est, X, y, train_idx)
+        X_te, _ = _safe_split(est, X, y, test_idx, train_idx)
+        est.fit(X_tr, y_tr)
+        p = est.predict_proba(X_te)
+        cols = np.searchsorted(classes, np.asarray(est.classes_))
+        full = np.zeros((p.shape[0], len(classes)))
+        full[:, cols] = p
+        return test_idx, full
 
     if sc is None:
         results = run_local_tasks(task_fn, folds)

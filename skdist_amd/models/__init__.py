@@ -11,6 +11,8 @@ weights only.
 """
 
 from .boosting import (
+    GradientBoostingClassifier,
+    GradientBoostingRegressor,
     HistGradientBoostingClassifier,
     HistGradientBoostingRegressor,
 )
@@ -19,4 +21,5 @@ from .linear import LinearSVC, LogisticRegression, Ridge
 __all__ = [
     "LogisticRegression", "LinearSVC", "Ridge",
     "HistGradientBoostingClassifier", "HistGradientBoostingRegressor",
+    "GradientBoostingClassifier", "GradientBoostingRegressor",
 ]

@@ -52,13 +52,23 @@ def _leaf_rows(tree, X):
 
 
 class _BaseHistGB(BaseEstimator):
+    """NAMING NOTE: despite the ``HistGradientBoosting*`` class names
+    (they ARE histogram-binned boosters), the parameter surface is
+    sklearn's ``GradientBoosting*`` API (``n_estimators``, ``subsample``,
+    ``max_depth=3`` semantics), NOT sklearn's HGB API
+    (``l2_regularization``, ``max_leaf_nodes``).  ``max_iter`` is accepted
+    as an alias for ``n_estimators`` so HGB-style grids still work; the
+    classes are also exported as ``GradientBoostingClassifier`` /
+    ``GradientBoostingRegressor``."""
+
     _is_classifier = False
 
     def __init__(self, n_estimators=100, learning_rate=0.1, max_depth=3,
                  subsample=1.0, min_samples_split=2, min_samples_leaf=1,
                  max_features=None, n_iter_no_change=None,
                  validation_fraction=0.1, tol=1e-4, random_state=None,
-                 sc=None):
+                 max_iter=None, sc=None):
+        self.max_iter = max_iter
         self.n_estimators = n_estimators
         self.learning_rate = learning_rate
         self.max_depth = max_depth
@@ -85,8 +95,14 @@ class _BaseHistGB(BaseEstimator):
             sc.device if sc is not None
             else ("cuda" if torch.cuda.is_available() else "cpu")
         )
-        rng = np.random.RandomState(
-            self.random_state if self.random_state is not None else 0
+        # sklearn semantics: random_state=None draws fresh entropy per
+        # fit (check_random_state), an int seed is deterministic
+        from sklearn.utils import check_random_state
+
+        rng = check_random_state(self.random_state)
+        n_rounds = (
+            int(self.max_iter) if self.max_iter is not None
+            else int(self.n_estimators)
         )
 
         F, K = self._init_raw(y, n)   # raw scores [n, K]; K columns
@@ -118,7 +134,7 @@ class _BaseHistGB(BaseEstimator):
         stall = 0
 
         stages = []
-        for _ in range(self.n_estimators):
+        for _ in range(n_rounds):
             if self.subsample < 1.0:
                 mask = (
                     rng.random_sample(n) < self.subsample
@@ -398,3 +414,9 @@ class HistGradientBoostingClassifier(ClassifierMixin, _BaseHistGB):
     def staged_decision_function(self, X):
         for r in self._staged_raw(X):
             yield r[:, 0] if r.shape[1] == 1 else r
+
+
+# honest-name aliases: the parameter surface IS sklearn's
+# GradientBoosting* API (see _BaseHistGB docstring)
+GradientBoostingClassifier = HistGradientBoostingClassifier
+GradientBoostingRegressor = HistGradientBoostingRegressor

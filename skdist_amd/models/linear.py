@@ -175,10 +175,14 @@ class _BatchedLinearBase(BaseEstimator):
 
         is_clf = isinstance(self, ClassifierMixin)
         t0 = time.perf_counter()
-        # class_weight folds into the row-weight plane; computed from the
-        # FULL y (documented deviation: sklearn recomputes 'balanced' per
-        # training fold — fold class ratios match the full data for
-        # stratified/partition CV)
+        # class_weight folds into the row-weight plane.  sklearn computes
+        # 'balanced' from each TRAINING fold's y; the plane is one weight
+        # per row shared by every column, so the batched path uses the
+        # full-y weights — exact for stratified folds, and guarded here:
+        # if any fold's class ratios deviate >1% from the full data the
+        # solve falls back to the generic per-task path (which matches
+        # sklearn exactly) instead of silently diverging.
+        self._check_balanced_foldable(y, cv_splits)
         sample_weight = self._merged_sample_weight(y, sample_weight)
         ds = DeviceDataset(
             X, y,
@@ -533,6 +537,36 @@ class _BatchedLinearBase(BaseEstimator):
         else:
             ests = [local[i] for i in range(len(problems))]
         return classes, ests
+
+    def _check_balanced_foldable(self, y, cv_splits):
+        """class_weight='balanced' is fold-dependent in sklearn; the
+        batched solve shares one row-weight plane, so it is only kept
+        for stratified folds: every training fold's class counts must be
+        within count-rounding (±1, +1% slack — what StratifiedKFold
+        guarantees) of proportional to the full data's.  Anything else
+        raises FallbackToGeneric (the generic per-task path recomputes
+        'balanced' per fold, matching sklearn exactly)."""
+        if getattr(self, "class_weight", None) != "balanced":
+            return
+        yv = np.asarray(y)
+        classes, full = np.unique(yv, return_counts=True)
+        n = len(yv)
+        for tr, _ in cv_splits:
+            cnt = np.array(
+                [(yv[tr] == c).sum() for c in classes], dtype=np.float64
+            )
+            if (cnt == 0).any():
+                raise FallbackToGeneric(
+                    "class_weight='balanced': a training fold misses a "
+                    "class; per-fold weights diverge — generic path"
+                )
+            expected = full * (len(tr) / n)
+            if np.max(np.abs(cnt - expected) - 0.01 * expected) > 1.0:
+                raise FallbackToGeneric(
+                    "class_weight='balanced' with non-stratified folds: "
+                    "per-fold weights diverge from full-data weights — "
+                    "generic path (exact sklearn semantics)"
+                )
 
     def _device_metric(self, scoring):
         if scoring is None:

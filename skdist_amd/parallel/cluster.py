@@ -190,18 +190,20 @@ class Cluster:
         mine = self.shard_indices(len(tasks))
         local = {}
         for i in mine:
-            if self.distributed:
-                try:
-                    local[i] = task_fn(tasks[i])
-                except Exception as e:  # noqa: BLE001 — re-raised post-gather
-                    import traceback
-
-                    local[i] = _TaskError(
-                        i, type(e).__name__, str(e),
-                        traceback.format_exc(),
-                    )
-            else:
+            # failures wrap identically in local and distributed mode so
+            # callers catching TaskFailedError behave the same at
+            # world_size 1 and N (in distributed mode the wrap also keeps
+            # this rank from crashing mid-collective and stranding the
+            # other ranks in the all-gather)
+            try:
                 local[i] = task_fn(tasks[i])
+            except Exception as e:  # noqa: BLE001 — re-raised post-gather
+                import traceback
+
+                local[i] = _TaskError(
+                    i, type(e).__name__, str(e),
+                    traceback.format_exc(),
+                )
         results = self.gather_task_results(local, len(tasks))
         for r in results:
             if isinstance(r, _TaskError):
