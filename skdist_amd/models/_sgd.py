@@ -306,6 +306,12 @@ def batched_sgd_fit(ds, spec, loss, epochs, batch_size, seed=0,
     the host RNG, applied as a device gather); batch composition stays
     fixed across epochs, matching the HIP path (hip_sgd_solve).
     """
+    if getattr(ds, "is_sparse", False):
+        from ._sparse_sgd import sparse_sgd_fit
+
+        return sparse_sgd_fit(ds, spec, loss, epochs, batch_size,
+                              seed=seed, momentum=momentum,
+                              lr_decay=lr_decay, force_eager=force_eager)
     device = ds.device
     n, fa = ds.Xaug.shape
     loss_id = _LOSS_IDS[loss] if isinstance(loss, str) else loss
@@ -417,6 +423,11 @@ def batched_scores_by_fold(ds, W, model_folds, col_class, n_classes,
     target class (n_classes consecutive columns per model when k > 2).
     Returns np.ndarray [n_models].
     """
+    if getattr(ds, "is_sparse", False):
+        from ._sparse_sgd import sparse_scores_by_fold
+
+        return sparse_scores_by_fold(ds, W, model_folds, col_class,
+                                     n_classes, metric)
     device = ds.device
     cpm = n_classes if n_classes > 2 else 1
     n_models = len(model_folds)

@@ -34,6 +34,34 @@ class FallbackToGeneric(Exception):
     falls back to the per-task generic path."""
 
 
+def _make_dataset(X, y, cluster=None, device=None, standardize=True,
+                  sample_weight=None, task=None, classes=None):
+    """DeviceDataset (dense MFMA path) or SparseDeviceDataset (CSR
+    text-scale path).  Sparse X stays sparse when densifying would be
+    unreasonable (> SKDIST_AMD_SPARSE_GB dense, default 2 GB, or wider
+    than 2^16 features — hashed-text shapes); smaller sparse inputs
+    densify onto the faster MFMA path as before.  SKDIST_AMD_FORCE_SPARSE=1
+    forces the sparse path (tests)."""
+    import os as _os
+
+    import scipy.sparse as _sp
+
+    if _sp.issparse(X):
+        force = _os.environ.get("SKDIST_AMD_FORCE_SPARSE") == "1"
+        gb = X.shape[0] * X.shape[1] * 4.0 / 1e9
+        wide = X.shape[1] > 65536
+        limit = float(_os.environ.get("SKDIST_AMD_SPARSE_GB", "2"))
+        if force or wide or gb > limit:
+            from ._sparse_sgd import SparseDeviceDataset
+
+            return SparseDeviceDataset(
+                X, y, cluster=cluster, device=device,
+                sample_weight=sample_weight, task=task, classes=classes)
+    return DeviceDataset(
+        X, y, cluster=cluster, device=device, standardize=standardize,
+        sample_weight=sample_weight, task=task, classes=classes)
+
+
 _DEVICE_METRICS = {
     "accuracy", "f1", "f1_weighted", "f1_macro", "neg_log_loss", "roc_auc",
     "r2", "neg_mean_squared_error",
@@ -74,7 +102,7 @@ class _BatchedLinearBase(BaseEstimator):
         sc = getattr(self, "sc", None)
         is_clf = isinstance(self, ClassifierMixin)
         sample_weight = self._merged_sample_weight(y, sample_weight)
-        ds = DeviceDataset(
+        ds = _make_dataset(
             X, y,
             cluster=None,  # single fit: local device, no broadcast
             device=sc.device if sc is not None else None,
@@ -184,7 +212,7 @@ class _BatchedLinearBase(BaseEstimator):
         # sklearn exactly) instead of silently diverging.
         self._check_balanced_foldable(y, cv_splits)
         sample_weight = self._merged_sample_weight(y, sample_weight)
-        ds = DeviceDataset(
+        ds = _make_dataset(
             X, y,
             cluster=cluster, standardize=self.standardize,
             sample_weight=sample_weight,
@@ -483,7 +511,7 @@ class _BatchedLinearBase(BaseEstimator):
                 "class_weight is per-binary-problem: generic path"
             )
         t0 = time.perf_counter()
-        ds = DeviceDataset(
+        ds = _make_dataset(
             X, y,
             cluster=cluster, standardize=self.standardize,
         )

@@ -18,7 +18,8 @@ HERE = os.path.dirname(os.path.abspath(__file__))
 CSRC = os.path.join(HERE, "csrc")
 OUT = os.path.join(HERE, "_skdist_hip.so")
 
-KERNEL_SOURCES = ["sgd_kernels.hip", "tree_kernels.hip", "predict_kernels.hip",
+KERNEL_SOURCES = ["sgd_kernels.hip", "sparse_sgd_kernels.hip",
+                  "tree_kernels.hip", "predict_kernels.hip",
                   "hash_kernels.hip"]
 BINDING_SOURCES = ["bindings.cpp"]
 

@@ -120,6 +120,120 @@ void sgd_epoch(torch::Tensor Xs, torch::Tensor XsT, torch::Tensor GT,
 }  // namespace
 
 // ------------------------------------------------------------------ //
+// sparse text-scale solver (sparse_sgd_kernels.hip)
+// ------------------------------------------------------------------ //
+
+extern "C" hipError_t skdist_sp_sgd_step(
+    const void* crow, const void* cidx, const void* cval,
+    void* W, void* Wb, void* s, void* G, void* part,
+    const void* y, const void* fold, const void* row_w,
+    const void* col_class, const void* col_fold, const void* col_class2,
+    const void* col_lr, const void* col_l2,
+    const void* ufeat, const void* cptr, const void* ridx,
+    const void* bval, long long uf,
+    long long start, long long m, int cp, int loss_id,
+    float inv_m, float lr_scale, hipStream_t stream);
+extern "C" hipError_t skdist_sp_forward(
+    const void* crow, const void* cidx, const void* cval,
+    const void* W, const void* Wb, const void* s, const void* rows,
+    void* Z, long long m, int cp, hipStream_t stream);
+extern "C" hipError_t skdist_sp_renorm(
+    void* W, const void* s, long long f, int cp, hipStream_t stream);
+
+namespace {
+
+static void check_cp(int64_t cp) {
+    TORCH_CHECK(cp % 64 == 0 && (cp <= 256 || cp % 256 == 0),
+                "cp must be a multiple of 64 (<=256) or of 256");
+}
+
+// one epoch of the sparse solver: the per-batch loop runs in C++.
+// CSC arrays are the concatenation over batches; ub_ptr (CPU int64,
+// [n_batches+1]) slices ufeat/cptr per batch, cptr values are GLOBAL
+// offsets into ridx/bval.
+void sp_sgd_epoch(torch::Tensor crow, torch::Tensor cidx,
+                  torch::Tensor cval, torch::Tensor W, torch::Tensor Wb,
+                  torch::Tensor s, torch::Tensor G, torch::Tensor part,
+                  torch::Tensor y, torch::Tensor fold, torch::Tensor row_w,
+                  torch::Tensor col_class, torch::Tensor col_fold,
+                  torch::Tensor col_class2, torch::Tensor col_lr,
+                  torch::Tensor col_l2, torch::Tensor ufeat,
+                  torch::Tensor cptr, torch::Tensor ridx,
+                  torch::Tensor bval, torch::Tensor ub_ptr_cpu,
+                  torch::Tensor inv_m_cpu, int64_t batch_size,
+                  int64_t loss_id, double lr_scale) {
+    for (auto* t : {&crow, &cidx, &cval, &W, &Wb, &s, &G, &part, &y,
+                    &fold, &col_class, &col_fold, &col_class2, &col_lr,
+                    &col_l2, &ufeat, &cptr, &ridx, &bval}) {
+        CHECK_DEV(*t);
+        CHECK_CONT(*t);
+    }
+    TORCH_CHECK(!ub_ptr_cpu.is_cuda() && !inv_m_cpu.is_cuda(),
+                "ub_ptr/inv_m must be CPU tensors");
+    TORCH_CHECK(crow.scalar_type() == torch::kInt64, "crow must be i64");
+    TORCH_CHECK(cidx.scalar_type() == torch::kInt32, "cidx must be i32");
+    TORCH_CHECK(G.scalar_type() == torch::kBFloat16, "G must be bf16");
+    const int64_t cp = W.size(1);
+    check_cp(cp);
+    const int64_t n = crow.size(0) - 1;
+    TORCH_CHECK(G.size(0) >= std::min<int64_t>(batch_size, n) &&
+                G.size(1) == cp, "G buffer too small");
+    const int64_t* ub = (const int64_t*)ub_ptr_cpu.data_ptr();
+    const float* inv_p = (const float*)inv_m_cpu.data_ptr();
+    auto stream = c10::hip::getCurrentHIPStream().stream();
+    int64_t bi = 0;
+    for (int64_t start = 0; start < n; start += batch_size, ++bi) {
+        const int64_t m = std::min(batch_size, n - start);
+        const int64_t ub0 = ub[bi], uf = ub[bi + 1] - ub[bi];
+        hipError_t err = skdist_sp_sgd_step(
+            crow.data_ptr(), cidx.data_ptr(), cval.data_ptr(),
+            W.data_ptr(), Wb.data_ptr(), s.data_ptr(), G.data_ptr(),
+            part.data_ptr(), y.data_ptr(), fold.data_ptr(),
+            row_w.numel() ? row_w.data_ptr() : nullptr,
+            col_class.data_ptr(), col_fold.data_ptr(),
+            col_class2.data_ptr(), col_lr.data_ptr(), col_l2.data_ptr(),
+            (const int*)ufeat.data_ptr() + ub0,
+            (const long long*)cptr.data_ptr() + ub0,
+            ridx.data_ptr(), bval.data_ptr(), uf,
+            start, m, (int)cp, (int)loss_id, inv_p[bi],
+            (float)lr_scale, stream);
+        TORCH_CHECK(err == hipSuccess, "sp_sgd_epoch: ",
+                    hipGetErrorString(err));
+    }
+}
+
+void sp_forward(torch::Tensor crow, torch::Tensor cidx,
+                torch::Tensor cval, torch::Tensor W, torch::Tensor Wb,
+                torch::Tensor s, torch::Tensor rows, torch::Tensor Z) {
+    for (auto* t : {&crow, &cidx, &cval, &W, &Wb, &s, &rows, &Z}) {
+        CHECK_DEV(*t);
+        CHECK_CONT(*t);
+    }
+    TORCH_CHECK(rows.scalar_type() == torch::kInt64, "rows must be i64");
+    const int64_t cp = W.size(1);
+    check_cp(cp);
+    TORCH_CHECK(Z.size(0) == rows.size(0) && Z.size(1) == cp,
+                "Z shape mismatch");
+    auto stream = c10::hip::getCurrentHIPStream().stream();
+    hipError_t err = skdist_sp_forward(
+        crow.data_ptr(), cidx.data_ptr(), cval.data_ptr(), W.data_ptr(),
+        Wb.data_ptr(), s.data_ptr(), rows.data_ptr(), Z.data_ptr(),
+        rows.size(0), (int)cp, stream);
+    TORCH_CHECK(err == hipSuccess, "sp_forward: ", hipGetErrorString(err));
+}
+
+void sp_renorm(torch::Tensor W, torch::Tensor s) {
+    CHECK_DEV(W);
+    CHECK_CONT(W);
+    auto stream = c10::hip::getCurrentHIPStream().stream();
+    hipError_t err = skdist_sp_renorm(W.data_ptr(), s.data_ptr(),
+                                      W.size(0), (int)W.size(1), stream);
+    TORCH_CHECK(err == hipSuccess, "sp_renorm: ", hipGetErrorString(err));
+}
+
+}  // namespace
+
+// ------------------------------------------------------------------ //
 // histogram tree builder (tree_kernels.hip)
 // ------------------------------------------------------------------ //
 
@@ -323,6 +437,10 @@ void hash_vectorize(torch::Tensor bytes, torch::Tensor doc_off,
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("sgd_step", &sgd_step, "fused batched SGD step (K1+K2+K3)");
     m.def("sgd_epoch", &sgd_epoch, "one epoch of fused SGD steps");
+    m.def("sp_sgd_epoch", &sp_sgd_epoch,
+          "one epoch of the sparse text-scale solver");
+    m.def("sp_forward", &sp_forward, "sparse decision values (scoring)");
+    m.def("sp_renorm", &sp_renorm, "fold the lazy-L2 scale into W");
     m.def("score_fold", &score_fold,
           "fused fold-scoring GEMM + per-column stats");
     m.def("standardize", &standardize,

@@ -92,10 +92,18 @@ def test_ovr_ovo_sparse(sparse_xy):
     assert (blob.predict(Xs) == ovr.predict(Xs)).all()
 
 
-def test_densify_size_guard():
-    big = sp.csr_matrix((10_000, 3_000_000), dtype=np.float32)
-    with pytest.raises(ValueError, match="densify"):
-        LogisticRegression(epochs=1).fit(big, np.zeros(10_000, np.int64))
+def test_wide_sparse_routes_to_sparse_native_path():
+    """Round 1 raised a densify-size ValueError here; round 2's sparse-
+    native solver (models/_sparse_sgd.py) fits this shape directly."""
+    rng = np.random.default_rng(0)
+    rows = np.repeat(np.arange(10_000), 5)
+    cols = rng.integers(0, 3_000_000, size=len(rows))
+    vals = rng.standard_normal(len(rows)).astype(np.float32)
+    big = sp.csr_matrix((vals, (rows, cols)), shape=(10_000, 3_000_000))
+    y = (np.asarray(big[:, :100_000].sum(axis=1)).ravel() > 0).astype(int)
+    m = LogisticRegression(epochs=2, momentum=0.0).fit(big, y)
+    assert m.coef_.shape == (1, 3_000_000)
+    assert (m.predict(big) == y).mean() > 0.6
 
 
 def test_text_pipeline_sparse_end_to_end():
