@@ -45,11 +45,9 @@ LOSS_LOG, LOSS_HINGE, LOSS_SQUARED = 0, 1, 2
 
 
 def _pad_cols(c):
-    """Column padding rule shared with the kernels: multiples of 64 up
-    to 256, multiples of 256 beyond."""
-    if c <= 256:
-        return (c + 63) // 64 * 64
-    return (c + 255) // 256 * 256
+    """Column padding rule shared with the kernels (multiple of 64; the
+    kernels bounds-guard the tail column tile)."""
+    return (c + 63) // 64 * 64
 
 
 class SparseDeviceDataset:

@@ -143,8 +143,7 @@ extern "C" hipError_t skdist_sp_renorm(
 namespace {
 
 static void check_cp(int64_t cp) {
-    TORCH_CHECK(cp % 64 == 0 && (cp <= 256 || cp % 256 == 0),
-                "cp must be a multiple of 64 (<=256) or of 256");
+    TORCH_CHECK(cp % 64 == 0, "cp must be a multiple of 64");
 }
 
 // one epoch of the sparse solver: the per-batch loop runs in C++.

@@ -75,7 +75,7 @@ extern "C" __global__ __launch_bounds__(256) void k_sp_fwd(
     const int rpb = 256 / tc;
     const long long ri = (long long)blockIdx.x * rpb + threadIdx.x / tc;
     const int c = blockIdx.y * tc + (threadIdx.x % tc);
-    if (ri >= m) return;
+    if (ri >= m || c >= cp) return;
     const long long r = (rows != nullptr) ? rows[ri] : start + ri;
 
     const long long k0 = crow[r], k1 = crow[r + 1];
@@ -165,7 +165,7 @@ extern "C" __global__ __launch_bounds__(256) void k_sp_update(
     const int fpb = 256 / tc;
     const long long si = (long long)blockIdx.x * fpb + threadIdx.x / tc;
     const int c = blockIdx.y * tc + (threadIdx.x % tc);
-    if (si >= uf) return;
+    if (si >= uf || c >= cp) return;
     const int j = ufeat[si];
     const long long k0 = cptr[si], k1 = cptr[si + 1];
     float a0 = 0.f, a1 = 0.f, a2 = 0.f, a3 = 0.f;
@@ -206,7 +206,8 @@ static inline void sp_fwd_launch(
 {
     const int tc = cp >= 256 ? 256 : cp;
     const int rpb = 256 / tc;
-    dim3 grid((unsigned)((m + rpb - 1) / rpb), (unsigned)(cp / tc));
+    dim3 grid((unsigned)((m + rpb - 1) / rpb),
+              (unsigned)((cp + tc - 1) / tc));
     hipLaunchKernelGGL(k_sp_fwd, grid, dim3(256), 0, st,
                        (const long long*)crow, (const int*)cidx,
                        (const float*)cval, (const float*)W,
@@ -250,7 +251,8 @@ extern "C" hipError_t skdist_sp_sgd_step(
     {
         const int tc = cp >= 256 ? 256 : cp;
         const int fpb = 256 / tc;
-        dim3 grid((unsigned)((uf + fpb - 1) / fpb), (unsigned)(cp / tc));
+        dim3 grid((unsigned)((uf + fpb - 1) / fpb),
+                  (unsigned)((cp + tc - 1) / tc));
         hipLaunchKernelGGL(k_sp_update, grid, dim3(256), 0, stream,
                            (const int*)ufeat, (const long long*)cptr,
                            (const int*)ridx, (const float*)bval,
