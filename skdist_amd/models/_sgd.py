@@ -299,7 +299,8 @@ class ColumnSpec:
 
 
 def batched_sgd_fit(ds, spec, loss, epochs, batch_size, seed=0,
-                    momentum=0.0, lr_decay=0.0, force_eager=False):
+                    momentum=0.0, lr_decay=0.0, force_eager=False,
+                    adaptive=None):
     """Train all columns; returns W [f+1, ncols] fp32 on ds.device.
 
     Mini-batches walk ONE fixed seeded permutation of the rows (drawn on
@@ -311,7 +312,8 @@ def batched_sgd_fit(ds, spec, loss, epochs, batch_size, seed=0,
 
         return sparse_sgd_fit(ds, spec, loss, epochs, batch_size,
                               seed=seed, momentum=momentum,
-                              lr_decay=lr_decay, force_eager=force_eager)
+                              lr_decay=lr_decay, force_eager=force_eager,
+                              adaptive=adaptive)
     device = ds.device
     n, fa = ds.Xaug.shape
     loss_id = _LOSS_IDS[loss] if isinstance(loss, str) else loss

@@ -243,11 +243,18 @@ class SparseDeviceDataset:
 # --------------------------------------------------------------------- #
 
 def sparse_sgd_fit(ds, spec, loss, epochs, batch_size, seed=0,
-                   momentum=0.0, lr_decay=0.0, force_eager=False):
+                   momentum=0.0, lr_decay=0.0, force_eager=False,
+                   adaptive=None):
     """Train all columns on the sparse dataset; returns W [f+1, ncols]
     fp32 on ds.device (row f = intercept), same contract as
     ``batched_sgd_fit``.  ``momentum`` > 0 is dropped with a one-time
-    warning (module docstring)."""
+    warning (module docstring).
+
+    ``adaptive`` (default True): Adagrad-normalize the data-gradient
+    step per (feature, column) — rare text features take full-size
+    first steps instead of 1/batch-size ones.  The lazy L2 decay is
+    unaffected.  ``adaptive=False`` runs the dense solver's plain-SGD
+    update (the dense-equivalence mode the tests use)."""
     if momentum > 0.0:
         warnings.warn(
             "sparse batched solve runs momentum-free SGD (a momentum "
@@ -255,6 +262,8 @@ def sparse_sgd_fit(ds, spec, loss, epochs, batch_size, seed=0,
             "momentum=0 to silence", stacklevel=2)
     if getattr(spec, "feat_mask", None) is not None:
         raise ValueError("feat_mask is not supported on the sparse path")
+    if adaptive is None:
+        adaptive = True
     loss_id = _LOSS_IDS[loss] if isinstance(loss, str) else loss
     dev = ds.device
     ncols = spec.ncols
@@ -267,6 +276,12 @@ def sparse_sgd_fit(ds, spec, loss, epochs, batch_size, seed=0,
     W = torch.zeros(ds.f + 1, cp, dtype=torch.float32, device=dev)
     Wb = torch.zeros(cp, dtype=torch.float32, device=dev)
     s = torch.ones(cp, dtype=torch.float32, device=dev)
+    if adaptive:
+        H = torch.zeros(ds.f, cp, dtype=torch.float32, device=dev)
+        Hb = torch.zeros(cp, dtype=torch.float32, device=dev)
+    else:
+        H = torch.empty(0, dtype=torch.float32, device=dev)
+        Hb = torch.empty(0, dtype=torch.float32, device=dev)
 
     hip = (not force_eager) and _use_hip(dev)
     if hip:
@@ -284,8 +299,8 @@ def sparse_sgd_fit(ds, spec, loss, epochs, batch_size, seed=0,
         for epoch in range(epochs):
             lr_scale = 1.0 / (1.0 + lr_decay * epoch)
             ext.sp_sgd_epoch(
-                sh["crow"], sh["cidx"], sh["cval"], W[: ds.f], Wb, s, G,
-                part, sh["y"], sh["fold"], rw,
+                sh["crow"], sh["cidx"], sh["cval"], W[: ds.f], Wb, s,
+                H, Hb, G, part, sh["y"], sh["fold"], rw,
                 col["cls"], col["fold"], col["cls2"], col["lr"],
                 col["l2"], sh["ufeat"], sh["cptr"], sh["ridx"],
                 sh["bval"], sh["ub_ptr"], sh["inv_m"],
@@ -295,7 +310,7 @@ def sparse_sgd_fit(ds, spec, loss, epochs, batch_size, seed=0,
                 ext.sp_renorm(W[: ds.f], s)
                 s.fill_(1.0)
     else:
-        _sparse_sgd_eager(ds, sh, W, Wb, s, col, loss_id, epochs,
+        _sparse_sgd_eager(ds, sh, W, Wb, s, H, Hb, col, loss_id, epochs,
                           batch_size, lr_decay, cp)
 
     W[: ds.f].mul_(s.unsqueeze(0))
@@ -318,8 +333,8 @@ def _padded_cols(spec, cp, dev):
     }
 
 
-def _sparse_sgd_eager(ds, sh, W, Wb, s, col, loss_id, epochs, batch_size,
-                      lr_decay, cp):
+def _sparse_sgd_eager(ds, sh, W, Wb, s, H, Hb, col, loss_id, epochs,
+                      batch_size, lr_decay, cp):
     """Eager torch mirror of the kernels' number flow (fp32 math, bf16
     G round-trip, lazy L2 scale).  The numerics reference for the GPU
     tests; also the CPU execution path."""
@@ -377,11 +392,22 @@ def _sparse_sgd_eager(ds, sh, W, Wb, s, col, loss_id, epochs, batch_size,
 
             im = float(inv_m[bi])
             lr = col["lr"] * lr_scale
-            Wb.sub_(lr * G.sum(dim=0) * im)
+            gb = G.sum(dim=0) * im
+            if Hb.numel():
+                Hb.add_(gb * gb)
+                Wb.sub_(lr * gb * torch.rsqrt(Hb + 1e-12))
+            else:
+                Wb.sub_(lr * gb)
             s.mul_(1.0 - lr * col["l2"])
             GW = torch.zeros_like(Wt)
             GW.index_add_(0, jj, vv.unsqueeze(1) * G.index_select(0, rl))
-            Wt.sub_(lr.unsqueeze(0) * GW * im / s.unsqueeze(0))
+            GW.mul_(im)
+            if H.numel():
+                H.add_(GW * GW)
+                Wt.sub_(lr.unsqueeze(0) * GW * torch.rsqrt(H + 1e-12)
+                        / s.unsqueeze(0))
+            else:
+                Wt.sub_(lr.unsqueeze(0) * GW / s.unsqueeze(0))
         # renorm check per epoch, matching the HIP driver
         if float(s.min()) < 1e-3:
             Wt.mul_(s.unsqueeze(0))

@@ -133,6 +133,7 @@ class _BatchedLinearBase(BaseEstimator):
             ds, spec, self._loss, self.epochs, self.batch_size,
             seed=self._seed(), momentum=self.momentum,
             lr_decay=getattr(self, "lr_decay", 0.0),
+            adaptive=getattr(self, "adaptive", None),
         )
         self._store_fitted(ds, W, is_clf)
         self.n_features_in_ = ds.f
@@ -293,6 +294,7 @@ class _BatchedLinearBase(BaseEstimator):
                 ds, spec, self._loss, self.epochs, self.batch_size,
                 seed=self._seed(), momentum=self.momentum,
                 lr_decay=getattr(self, "lr_decay", 0.0),
+                adaptive=getattr(self, "adaptive", None),
             )
             fit_time = time.perf_counter() - t0
             t1 = time.perf_counter()
@@ -445,6 +447,7 @@ class _BatchedLinearBase(BaseEstimator):
                 ds, spec, self._loss, self.epochs, self.batch_size,
                 seed=self._seed(), momentum=self.momentum,
                 lr_decay=getattr(self, "lr_decay", 0.0),
+                adaptive=getattr(self, "adaptive", None),
             )
             scores = batched_scores_by_fold(
                 ds, W, np.asarray(model_folds),
@@ -551,6 +554,7 @@ class _BatchedLinearBase(BaseEstimator):
                 ds, spec, self._loss, self.epochs, self.batch_size,
                 seed=self._seed(), momentum=self.momentum,
                 lr_decay=getattr(self, "lr_decay", 0.0),
+                adaptive=getattr(self, "adaptive", None),
             )
             per = (time.perf_counter() - t0) / len(ids)
             for ci, pi in enumerate(ids):
@@ -643,8 +647,10 @@ class LogisticRegression(ClassifierMixin, _BatchedLinearBase):
 
     def __init__(self, C=1.0, lr=0.5, epochs=20, batch_size=8192,
                  momentum=0.9, lr_decay=0.0, standardize=True,
-                 class_weight=None, random_state=None, sc=None):
+                 class_weight=None, random_state=None, adaptive=None,
+                 sc=None):
         self.C = C
+        self.adaptive = adaptive
         self.class_weight = class_weight
         self.lr_decay = lr_decay
         self.lr = lr
@@ -681,8 +687,10 @@ class LinearSVC(ClassifierMixin, _BatchedLinearBase):
 
     def __init__(self, C=1.0, lr=0.5, epochs=20, batch_size=8192,
                  momentum=0.9, lr_decay=0.0, standardize=True,
-                 class_weight=None, random_state=None, sc=None):
+                 class_weight=None, random_state=None, adaptive=None,
+                 sc=None):
         self.C = C
+        self.adaptive = adaptive
         self.class_weight = class_weight
         self.lr_decay = lr_decay
         self.lr = lr
@@ -713,8 +721,9 @@ class Ridge(RegressorMixin, _BatchedLinearBase):
 
     def __init__(self, alpha=1.0, lr=0.5, epochs=20, batch_size=8192,
                  momentum=0.0, lr_decay=0.0, standardize=True,
-                 random_state=None, sc=None):
+                 random_state=None, adaptive=None, sc=None):
         self.alpha = alpha
+        self.adaptive = adaptive
         self.lr_decay = lr_decay
         self.lr = lr
         self.epochs = epochs

@@ -125,7 +125,7 @@ void sgd_epoch(torch::Tensor Xs, torch::Tensor XsT, torch::Tensor GT,
 
 extern "C" hipError_t skdist_sp_sgd_step(
     const void* crow, const void* cidx, const void* cval,
-    void* W, void* Wb, void* s, void* G, void* part,
+    void* W, void* Wb, void* s, void* h, void* hb, void* G, void* part,
     const void* y, const void* fold, const void* row_w,
     const void* col_class, const void* col_fold, const void* col_class2,
     const void* col_lr, const void* col_l2,
@@ -152,7 +152,8 @@ static void check_cp(int64_t cp) {
 // offsets into ridx/bval.
 void sp_sgd_epoch(torch::Tensor crow, torch::Tensor cidx,
                   torch::Tensor cval, torch::Tensor W, torch::Tensor Wb,
-                  torch::Tensor s, torch::Tensor G, torch::Tensor part,
+                  torch::Tensor s, torch::Tensor h, torch::Tensor hb,
+                  torch::Tensor G, torch::Tensor part,
                   torch::Tensor y, torch::Tensor fold, torch::Tensor row_w,
                   torch::Tensor col_class, torch::Tensor col_fold,
                   torch::Tensor col_class2, torch::Tensor col_lr,
@@ -186,7 +187,9 @@ void sp_sgd_epoch(torch::Tensor crow, torch::Tensor cidx,
         const int64_t ub0 = ub[bi], uf = ub[bi + 1] - ub[bi];
         hipError_t err = skdist_sp_sgd_step(
             crow.data_ptr(), cidx.data_ptr(), cval.data_ptr(),
-            W.data_ptr(), Wb.data_ptr(), s.data_ptr(), G.data_ptr(),
+            W.data_ptr(), Wb.data_ptr(), s.data_ptr(),
+            h.numel() ? h.data_ptr() : nullptr,
+            hb.numel() ? hb.data_ptr() : nullptr, G.data_ptr(),
             part.data_ptr(), y.data_ptr(), fold.data_ptr(),
             row_w.numel() ? row_w.data_ptr() : nullptr,
             col_class.data_ptr(), col_fold.data_ptr(),

@@ -129,10 +129,11 @@ extern "C" __global__ __launch_bounds__(256) void k_sp_colsum(
 }
 
 // K2b: reduce chunks -> intercept update + lazy-scale advance
-// grid: ceil(CP/256), block 256
+// grid: ceil(CP/256), block 256.  With `hb` non-null the intercept step
+// is Adagrad-normalized (hb accumulates the squared mean-gradient).
 extern "C" __global__ __launch_bounds__(256) void k_sp_bias_scale(
     const float* __restrict__ part, float* __restrict__ Wb,
-    float* __restrict__ s,
+    float* __restrict__ s, float* __restrict__ hb,
     const float* __restrict__ col_lr, const float* __restrict__ col_l2,
     int n_chunks, int cp, float inv_m, float lr_scale)
 {
@@ -141,14 +142,26 @@ extern "C" __global__ __launch_bounds__(256) void k_sp_bias_scale(
     float g = 0.f;
     for (int z = 0; z < n_chunks; ++z)
         g += part[(long long)z * cp + c];
+    g *= inv_m;
     const float lr = col_lr[c] * lr_scale;
-    Wb[c] -= lr * g * inv_m;                  // no L2 on the intercept
+    float step = lr * g;
+    if (hb != nullptr) {
+        const float h = hb[c] + g * g;
+        hb[c] = h;
+        step = lr * g * __frsqrt_rn(h + 1e-12f);
+    }
+    Wb[c] -= step;                            // no L2 on the intercept
     s[c] *= (1.f - lr * col_l2[c]);          // lazy decay of all of W
 }
 
 // ---------------------------------------------------------------------- //
 // K3: batch-CSC weight update — one (feature, column) per lane
 // grid: (ceil(uf / fpb), CP / TC), block 256
+// With `h` non-null the data-gradient step is Adagrad-normalized per
+// (feature, column): rare text features take full-size first steps
+// instead of 1/batch-size ones (the standard sparse-text optimizer);
+// the lazy L2 scale path is untouched, so regularization stays exactly
+// the dense solver's per-step decay.
 // ---------------------------------------------------------------------- //
 extern "C" __global__ __launch_bounds__(256) void k_sp_update(
     const int* __restrict__ ufeat,   // [uf]
@@ -157,6 +170,7 @@ extern "C" __global__ __launch_bounds__(256) void k_sp_update(
     const float* __restrict__ bval,
     const __bf16* __restrict__ G,    // [m][CP]
     float* __restrict__ W,           // [f][CP]
+    float* __restrict__ h,           // [f][CP] Adagrad accum, or null
     const float* __restrict__ s,     // [CP] (post-K2b value)
     const float* __restrict__ col_lr,
     long long uf, int cp, float inv_m, float lr_scale)
@@ -180,7 +194,13 @@ extern "C" __global__ __launch_bounds__(256) void k_sp_update(
         a0 += bval[k] * bf16_to_f32(G[(long long)ridx[k] * cp + c]);
     const float g = ((a0 + a1) + (a2 + a3)) * inv_m;
     const long long e = (long long)j * cp + c;
-    W[e] -= col_lr[c] * lr_scale * g / s[c];
+    float step = col_lr[c] * lr_scale * g;
+    if (h != nullptr) {
+        const float hv = h[e] + g * g;
+        h[e] = hv;
+        step *= __frsqrt_rn(hv + 1e-12f);
+    }
+    W[e] -= step / s[c];
 }
 
 // fold the scale into W when it drifts far from 1 (rare; keeps /s[c]
@@ -220,7 +240,7 @@ static inline void sp_fwd_launch(
 
 extern "C" hipError_t skdist_sp_sgd_step(
     const void* crow, const void* cidx, const void* cval,
-    void* W, void* Wb, void* s, void* G, void* part,
+    void* W, void* Wb, void* s, void* h, void* hb, void* G, void* part,
     const void* y, const void* fold, const void* row_w,
     const void* col_class, const void* col_fold, const void* col_class2,
     const void* col_lr, const void* col_l2,
@@ -244,6 +264,7 @@ extern "C" hipError_t skdist_sp_sgd_step(
         dim3 grid((unsigned)((cp + 255) / 256));
         hipLaunchKernelGGL(k_sp_bias_scale, grid, dim3(256), 0, stream,
                            (const float*)part, (float*)Wb, (float*)s,
+                           (float*)hb,
                            (const float*)col_lr, (const float*)col_l2,
                            nch, cp, inv_m, lr_scale);
         HIP_CHECK(hipGetLastError());
@@ -256,7 +277,8 @@ extern "C" hipError_t skdist_sp_sgd_step(
         hipLaunchKernelGGL(k_sp_update, grid, dim3(256), 0, stream,
                            (const int*)ufeat, (const long long*)cptr,
                            (const int*)ridx, (const float*)bval,
-                           (const __bf16*)G, (float*)W, (const float*)s,
+                           (const __bf16*)G, (float*)W, (float*)h,
+                           (const float*)s,
                            (const float*)col_lr, uf, cp, inv_m, lr_scale);
         HIP_CHECK(hipGetLastError());
     }

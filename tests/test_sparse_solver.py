@@ -37,7 +37,7 @@ def test_sparse_matches_dense_solver(force_sparse):
     from skdist_amd.models import LogisticRegression
 
     Xd, X, y, _ = _data()
-    kw = dict(epochs=20, momentum=0.0, random_state=0)
+    kw = dict(epochs=20, momentum=0.0, random_state=0, adaptive=False)
     m_sp = LogisticRegression(**kw).fit(X, y)
     os.environ.pop("SKDIST_AMD_FORCE_SPARSE")
     m_de = LogisticRegression(standardize=False, **kw).fit(Xd, y)
@@ -69,7 +69,7 @@ def test_sparse_search_scores_match_dense(force_sparse):
     Xd, X, y, _ = _data(n=2400, f=120, seed=1)
     cv = KFold(3)
     grid = {"C": [0.1, 1.0, 10.0]}
-    kw = dict(epochs=12, momentum=0.0, random_state=0)
+    kw = dict(epochs=12, momentum=0.0, random_state=0, adaptive=False)
     g_sp = DistGridSearchCV(
         LogisticRegression(**kw), grid, cv=cv).fit(X, y)
     os.environ.pop("SKDIST_AMD_FORCE_SPARSE")
@@ -134,7 +134,8 @@ def test_sparse_ridge_and_sample_weight(force_sparse):
 
     sw = rng.random(len(y)).astype(np.float32) + 0.5
     m_sp = LogisticRegression(
-        epochs=15, momentum=0.0, random_state=0).fit(X, y, sample_weight=sw)
+        epochs=15, momentum=0.0, random_state=0, adaptive=False
+    ).fit(X, y, sample_weight=sw)
     os.environ.pop("SKDIST_AMD_FORCE_SPARSE")
     m_de = LogisticRegression(
         epochs=15, momentum=0.0, standardize=False, random_state=0
@@ -148,7 +149,8 @@ def test_sparse_lazy_l2_matches_direct_decay(force_sparse):
     from skdist_amd.models import LogisticRegression
 
     Xd, X, y, _ = _data(n=1500, f=60, seed=6)
-    kw = dict(C=1e-3, epochs=20, lr=1.0, momentum=0.0, random_state=0)
+    kw = dict(C=1e-3, epochs=20, lr=1.0, momentum=0.0, random_state=0,
+              adaptive=False)
     m_sp = LogisticRegression(**kw).fit(X, y)
     os.environ.pop("SKDIST_AMD_FORCE_SPARSE")
     m_de = LogisticRegression(standardize=False, **kw).fit(Xd, y)
