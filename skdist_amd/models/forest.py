@@ -388,11 +388,28 @@ class ForestBuilder:
                                     device=ds.device)
                 w[t] = torch.bincount(idx, minlength=ds.n).clamp_(
                     max=255).to(torch.uint8)
+        self._w_scale = 1.0
         if sample_weight is not None:
             sw = torch.as_tensor(
-                np.ascontiguousarray(sample_weight), device=ds.device)
-            w = (w.to(torch.float32) * sw.unsqueeze(0)).round_().clamp_(
-                0, 255).to(torch.uint8)
+                np.ascontiguousarray(sample_weight, dtype=np.float32),
+                device=ds.device)
+            wf = w.to(torch.float32) * sw.unsqueeze(0)
+            mx = float(wf.max())
+            if mx <= 0:
+                raise ValueError("sample_weight leaves no weighted rows")
+            if mx <= 255 and bool((wf == wf.round()).all()):
+                # integer-valued weights (bootstrap counts, 0/1 masks,
+                # integer class weights): exact, as in round 1
+                w = wf.round_().to(torch.uint8)
+            else:
+                # real-valued weights quantize into the uint8 plane at
+                # 192 levels (~0.5% relative granularity).  A uniform
+                # weight scale is mathematically free for splits and
+                # leaf values; the min_samples/min-leaf thresholds are
+                # scaled to match (_build_batch).
+                self._w_scale = 192.0 / mx
+                w = (wf * self._w_scale).round_().clamp_(0, 255).to(
+                    torch.uint8)
         return w
 
     # -------------------------------------------------------------- #
@@ -403,6 +420,11 @@ class ForestBuilder:
         s_out = S if ds.is_cls else 1
 
         weights = self.make_weights(seeds, sample_weight)
+        # weighted-count thresholds live in the (possibly quantized)
+        # weight-plane scale
+        ws = getattr(self, "_w_scale", 1.0)
+        self._mss_eff = self.mss * ws
+        self._msl_eff = self.msl * ws
         si_a, counts = self._initial_sample_idx(weights)
         si_b = torch.empty_like(si_a)
 
@@ -438,7 +460,7 @@ class ForestBuilder:
             ok = (
                 (bfeat >= 0)
                 & (depth < self.max_depth)
-                & (wp >= self.mss)
+                & (wp >= self._mss_eff)
                 & (bgain > 0)
                 & ((wp / root_w[fr_tree]) * bgain >= self.mid - 1e-12)
             )
@@ -496,9 +518,9 @@ class ForestBuilder:
             st_r = st_l + nrows_l
 
             can_grow = depth + 1 < self.max_depth
-            grow_l = (can_grow & (wl >= self.mss) & (nrows_l > 1)
+            grow_l = (can_grow & (wl >= self._mss_eff) & (nrows_l > 1)
                       & (self._impurity_vec(ls, wl) > 1e-12))
-            grow_r = (can_grow & (wr >= self.mss) & (nrows_r > 1)
+            grow_r = (can_grow & (wr >= self._mss_eff) & (nrows_r > 1)
                       & (self._impurity_vec(rs, wr) > 1e-12))
             st.make_leaves(gl[~grow_l], self._leaf_values_vec(ls[~grow_l]))
             st.make_leaves(gr[~grow_r], self._leaf_values_vec(rs[~grow_r]))
@@ -728,7 +750,8 @@ class ForestBuilder:
         out_lstats = fbuf[(3 + S) * NF:].view(NF, S)
         self._ext.tree_split(
             hist, seed_t, ds.f, ds.nbins, S, int(ds.is_cls), self.crit,
-            self.m_features, int(self.extra_mode), float(self.msl),
+            self.m_features, int(self.extra_mode),
+            float(getattr(self, "_msl_eff", self.msl)),
             out_feat, out_bin, out_wl, out_gain, out_imp, out_stats,
             out_lstats)
         ih = ibuf.cpu().numpy()
@@ -854,7 +877,8 @@ class ForestBuilder:
                     (parent[2] - cum[:, :, 2]) / wr - mr * mr, 0.0)
         with np.errstate(invalid="ignore"):
             gain = imp_p - (wl * imp_l + wr * imp_r) / max(wp, 1e-30)
-        valid = (wl >= self.msl) & (wr >= self.msl) & sel[:, None]
+        msl = getattr(self, "_msl_eff", self.msl)
+        valid = (wl >= msl) & (wr >= msl) & sel[:, None]
         if self.extra_mode:
             extra_ok = np.zeros_like(valid)
             wbin = h.sum(axis=2) if ds.is_cls else h[:, :, 0]

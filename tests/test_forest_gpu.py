@@ -196,3 +196,56 @@ def test_subtraction_trick_identical_trees():
         np.testing.assert_array_equal(ta.threshold, tb.threshold)
         np.testing.assert_array_equal(ta.left, tb.left)
         np.testing.assert_allclose(ta.value, tb.value, atol=1e-6)
+
+
+def test_device_forest_sample_weight_matches_eager(monkeypatch):
+    """Real-valued sample_weight through the quantized uint8 plane: HIP
+    builder vs the eager mirror must produce identical trees."""
+    monkeypatch.setenv("SKDIST_AMD_ALLOW_EAGER", "1")
+    from skdist_amd.models.forest import BinnedDataset, ForestBuilder
+
+    X, y = _cls_data(n=6000, f=10, seed=11)
+    rng = np.random.default_rng(0)
+    sw = rng.random(len(y)) + 0.5
+
+    def build(engine):
+        ds = BinnedDataset(X, y, "cuda", is_cls=True)
+        b = ForestBuilder(ds, "gini", max_depth=7, min_samples_split=4,
+                          min_samples_leaf=2, max_features=None,
+                          extra_mode=False, bootstrap=True,
+                          engine=engine)
+        return b.build([5, 6], sample_weight=sw)
+
+    t_hip = build("hip")
+    t_eag = build("eager")
+    for a, b in zip(t_hip, t_eag):
+        np.testing.assert_array_equal(a.feature, b.feature)
+        np.testing.assert_array_equal(a.left, b.left)
+        np.testing.assert_allclose(a.value, b.value, atol=1e-5)
+
+
+def test_dist_forest_class_weight_device_path():
+    """class_weight='balanced' stays on the device path and shifts
+    minority-class recall up (round-1: silent CPU fallback)."""
+    from sklearn.metrics import recall_score
+
+    from skdist_amd import Cluster
+    from skdist_amd.distribute.ensemble import DistRandomForestClassifier
+
+    rng = np.random.default_rng(0)
+    n = 60_000
+    X = rng.standard_normal((n, 16)).astype(np.float32)
+    y = ((X[:, 0] * 2 - 2.6 + 0.5 * rng.standard_normal(n)) > 0).astype(
+        np.int64)
+    sc = Cluster(require_gpu=True)
+    m0 = DistRandomForestClassifier(
+        n_estimators=20, max_depth=6, random_state=0, sc=sc).fit(X, y)
+    sc2 = Cluster(require_gpu=True)
+    m1 = DistRandomForestClassifier(
+        n_estimators=20, max_depth=6, class_weight="balanced",
+        random_state=0, sc=sc2)
+    assert m1._device_fit_ok(sc2, X, None)  # no CPU fallback
+    m1.fit(X, y)
+    r0 = recall_score(y, m0.predict(X))
+    r1 = recall_score(y, m1.predict(X))
+    assert r1 > r0
