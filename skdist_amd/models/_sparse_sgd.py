@@ -290,8 +290,8 @@ def sparse_sgd_fit(ds, spec, loss, epochs, batch_size, seed=0,
         ext = require_hip()
         m0 = min(batch_size, ds.n)
         G = torch.empty(m0, cp, dtype=torch.bfloat16, device=dev)
-        part = torch.empty((m0 + 1023) // 1024, cp, dtype=torch.float32,
-                           device=dev)
+        # k_sp_colsum chunking caps n_chunks at 256 for any batch size
+        part = torch.empty(256, cp, dtype=torch.float32, device=dev)
         rw = (
             sh["rw"] if sh["rw"] is not None
             else torch.empty(0, dtype=torch.float32, device=dev)
