@@ -234,6 +234,10 @@ class _BatchedLinearBase(BaseEstimator):
                 np.array([1], dtype=np.int32) if k == 2
                 else np.arange(k, dtype=np.int32)
             )
+            if metric == "roc_auc" and k > 2:
+                raise FallbackToGeneric(
+                    "multiclass roc_auc needs sklearn's multi_class "
+                    "semantics: generic path")
         else:
             n_classes = 2
             cols_per_model = 1
@@ -395,6 +399,10 @@ class _BatchedLinearBase(BaseEstimator):
                 np.array([1], dtype=np.int32) if k == 2
                 else np.arange(k, dtype=np.int32)
             )
+            if metric == "roc_auc" and k > 2:
+                raise FallbackToGeneric(
+                    "multiclass roc_auc needs sklearn's multi_class "
+                    "semantics: generic path")
         else:
             n_classes = 2
             cls = np.array([-1], dtype=np.int32)
