@@ -49,6 +49,10 @@ def main():
     ap.add_argument("--features", type=int, default=256)
     ap.add_argument("--epochs", type=int, default=10)
     ap.add_argument("--batch-size", type=int, default=8192)
+    ap.add_argument("--dump-cv", default=None,
+                    help="rank 0 saves the last step's mean_test_score "
+                         "+ best_index_ to this .npz (dry-run equality "
+                         "checks across world sizes)")
     args = ap.parse_args()
 
     import torch
@@ -103,6 +107,13 @@ def main():
         elapsed = float(t.item())
 
     value = n_fits * args.steps / elapsed
+    if rank == 0 and args.dump_cv and gs is not None:
+        np.savez(
+            args.dump_cv,
+            mean_test_score=gs.cv_results_["mean_test_score"],
+            best_index=np.int64(gs.best_index_),
+            coef=gs.best_estimator_.coef_,
+        )
     if rank == 0:
         out = {
             "metric": "candidate-fits/sec",
