@@ -20,6 +20,7 @@ from ..preprocessing import (
     ImputeNull,
     MultihotEncoder,
     SelectField,
+    UniqueFactorizedVectorizer,
 )
 
 
@@ -47,10 +48,13 @@ def onehot_encoder(col):
             ("var", SelectField(cols=[col], single_dimension=True)),
             ("cast", FeatureCast(cast_type=str)),
             ("fillna", ImputeNull("")),
-            ("vec", CountVectorizer(
+            # CountVectorizer on the column's UNIQUE values only +
+            # row gather: sklearn-identical output, O(n_unique) python
+            # instead of O(n) (UniqueFactorizedVectorizer docstring)
+            ("vec", UniqueFactorizedVectorizer(CountVectorizer(
                 token_pattern=None, tokenizer=tokenizer, binary=True,
                 decode_error="ignore",
-            )),
+            ))),
         ]),
     )]
 
@@ -61,7 +65,7 @@ def multihot_encoder(col):
         Pipeline([
             ("var", SelectField(cols=[col], single_dimension=True)),
             ("fillna", ImputeNull([])),
-            ("vec", MultihotEncoder()),
+            ("vec", UniqueFactorizedVectorizer(MultihotEncoder())),
         ]),
     )]
 

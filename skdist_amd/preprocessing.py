@@ -12,7 +12,7 @@ import numpy as np
 import pandas as pd
 import scipy.sparse as sparse
 from sklearn import feature_selection
-from sklearn.base import BaseEstimator, TransformerMixin
+from sklearn.base import BaseEstimator, TransformerMixin, clone
 from sklearn.feature_extraction.text import HashingVectorizer
 from sklearn.preprocessing import LabelEncoder, MultiLabelBinarizer, normalize
 
@@ -20,6 +20,7 @@ __all__ = [
     "SelectField", "DenseTransformer", "SparseTransformer", "FeatureCast",
     "ImputeNull", "LabelEncoderPipe", "SelectorMem",
     "HashingVectorizerChunked", "MultihotEncoder",
+    "UniqueFactorizedVectorizer",
 ]
 
 
@@ -229,3 +230,73 @@ class MultihotEncoder(TransformerMixin, BaseEstimator):
             warnings.simplefilter("ignore")
             out = self.transformer.transform(X)
         return sparse.csr_matrix(out) if self.sparse_output else out
+
+
+class UniqueFactorizedVectorizer(TransformerMixin, BaseEstimator):
+    """Run a per-value vectorizer on the column's UNIQUE values only,
+    then expand rows by code gather — categorical columns have few
+    uniques, so the O(n) python-loop cost of sklearn vectorizers
+    (CountVectorizer's per-doc analyze, MultiLabelBinarizer) drops to
+    O(n_unique) + one vectorized gather (VERDICT round-2 item 8: wide
+    categorical frames were host-python-bound).
+
+    Exactness: the inner vectorizer IS the fitted estimator, fit on the
+    deduplicated values — its vocabulary is the same token set it would
+    build from the full column, so outputs are sklearn-identical.
+    (Only df-threshold options like ``min_df``/``max_df``/``max_features``
+    would see different document counts; the Encoderizer tiers use none
+    of them.)  Unhashable values (lists) are keyed by ``tuple``;
+    transform-time unseen values flow through the fitted inner
+    vectorizer exactly like any unseen document.
+    """
+
+    def __init__(self, inner):
+        self.inner = inner
+
+    @staticmethod
+    def _factorize(X):
+        vals = list(X)
+        keys = [
+            tuple(v) if isinstance(v, (list, set)) else v for v in vals
+        ]
+        seen = {}
+        codes = np.empty(len(keys), dtype=np.int64)
+        uniques = []
+        for i, k in enumerate(keys):
+            j = seen.get(k)
+            if j is None:
+                j = len(uniques)
+                seen[k] = j
+                uniques.append(vals[i])
+            codes[i] = j
+        return codes, uniques
+
+    def fit(self, X, y=None):
+        codes, uniques = self._factorize(X)
+        self.inner_ = clone(self.inner).fit(uniques)
+        return self
+
+    def transform(self, X, y=None):
+        codes, uniques = self._factorize(X)
+        M = self.inner_.transform(uniques)
+        if sparse.issparse(M):
+            return M.tocsr()[codes]
+        return np.asarray(M)[codes]
+
+    def fit_transform(self, X, y=None, **kw):
+        codes, uniques = self._factorize(X)
+        self.inner_ = clone(self.inner)
+        M = self.inner_.fit_transform(uniques)
+        if sparse.issparse(M):
+            return M.tocsr()[codes]
+        return np.asarray(M)[codes]
+
+    def __getattr__(self, name):
+        # expose the fitted inner vectorizer's attributes
+        # (vocabulary_, get_feature_names_out, classes_, ...)
+        if name.startswith("__") or name in ("inner", "inner_"):
+            raise AttributeError(name)
+        inner = self.__dict__.get("inner_")
+        if inner is None:
+            raise AttributeError(name)
+        return getattr(inner, name)
