@@ -111,7 +111,10 @@ class SparseDeviceDataset:
             yt = cluster.bcast_tensor(
                 torch.as_tensor(y_host).to(dev)
                 if y_host is not None else None)
-            kind, cls_arr = cluster.bcast_obj((kind, cls_arr))
+            kind, cls_arr, shape = cluster.bcast_obj(
+                (kind, cls_arr, getattr(self, "_shape", None)))
+            if shape is not None:
+                self._shape = shape
         else:
             crow = crow.to(self.device)
             cidx = cidx.to(self.device)
