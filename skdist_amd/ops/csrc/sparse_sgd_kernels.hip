@@ -150,10 +150,17 @@ extern "C" __global__ __launch_bounds__(256) void k_sp_bias_scale(
 {
     const int c = blockIdx.x * 256 + threadIdx.x;
     if (c >= cp) return;
-    float g = 0.f;
-    for (int z = 0; z < n_chunks; ++z)
-        g += part[(long long)z * cp + c];
-    g *= inv_m;
+    float g0 = 0.f, g1 = 0.f, g2 = 0.f, g3 = 0.f;
+    int z = 0;
+    for (; z + 4 <= n_chunks; z += 4) {
+        g0 += part[(long long)z * cp + c];
+        g1 += part[(long long)(z + 1) * cp + c];
+        g2 += part[(long long)(z + 2) * cp + c];
+        g3 += part[(long long)(z + 3) * cp + c];
+    }
+    for (; z < n_chunks; ++z)
+        g0 += part[(long long)z * cp + c];
+    float g = ((g0 + g1) + (g2 + g3)) * inv_m;
     const float lr = col_lr[c] * lr_scale;
     float step = lr * g;
     if (hb != nullptr) {

@@ -253,12 +253,17 @@ def test_dist_forest_class_weight_device_path():
     sc = Cluster(require_gpu=True)
     m0 = DistRandomForestClassifier(
         n_estimators=20, max_depth=6, random_state=0, sc=sc).fit(X, y)
+    import warnings as _w
+
     sc2 = Cluster(require_gpu=True)
     m1 = DistRandomForestClassifier(
         n_estimators=20, max_depth=6, class_weight="balanced",
         random_state=0, sc=sc2)
-    assert m1._device_fit_ok(sc2, X, None)  # no CPU fallback
-    m1.fit(X, y)
+    with _w.catch_warnings(record=True) as rec:
+        _w.simplefilter("always")
+        m1.fit(X, y)
+    # stayed on the device path: no CPU-fallback warning fired
+    assert not any("falling back" in str(r.message) for r in rec)
     r0 = recall_score(y, m0.predict(X))
     r1 = recall_score(y, m1.predict(X))
     assert r1 > r0

@@ -40,6 +40,8 @@ for db in "$PROF"/boost/*/*.db; do
   python tools/prof_summary.py "$db" > gpurun_out/r3_boost_summary.txt 2>&1
 done
 ls "$PROF"/sgd_pmc/ > gpurun_out/r3_pmc_ls.txt 2>&1
-python tools/pmc_summary.py "$PROF/sgd_pmc" > gpurun_out/r3_sgd_pmc_summary.txt 2>&1
+for db in "$PROF"/sgd_pmc/*/*.db; do
+  python tools/prof_summary.py "$db" > gpurun_out/r3_sgd_pmc_summary.txt 2>&1
+done
 du -sh gpurun_out >> gpurun_out/r3_pmc_ls.txt
 echo "ALL DONE rc=$?"
