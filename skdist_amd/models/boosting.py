@@ -51,6 +51,43 @@ def _leaf_rows(tree, X):
     return tree.left[tree.apply(X)]
 
 
+def _leaf_value_rows(tree, X, Xdev, dev):
+    """Device int64 [n] of each row's leaf-payload index.  On GPU this
+    is one k_forest_apply traversal of the RESIDENT X (no host
+    round-trip); on CPU the numpy traversal."""
+    if Xdev is None:
+        return torch.as_tensor(_leaf_rows(tree, X), dtype=torch.int64,
+                               device=dev)
+    from ..ops import require_hip
+
+    ext = require_hip()
+    as_t = lambda a, dt: torch.as_tensor(
+        np.ascontiguousarray(a), dtype=dt, device=dev)
+    feat = as_t(tree.feature, torch.int32)
+    thr = as_t(tree.threshold, torch.float32)
+    left = as_t(tree.left, torch.int32)
+    right = as_t(tree.right, torch.int32)
+    roots = torch.zeros(1, dtype=torch.int32, device=dev)
+    out = torch.empty(len(Xdev), 1, dtype=torch.int32, device=dev)
+    ext.forest_apply(Xdev, feat, thr, left, right, roots, out)
+    return left.index_select(
+        0, out[:, 0].to(torch.int64)).to(torch.int64)
+
+
+def _seg_sum(vrow, vals, nl):
+    """Deterministic per-leaf sums: stable sort by leaf + fp64 cumsum
+    differencing (GPU scatter_add atomics would be order-dependent)."""
+    dev = vals.device
+    order = torch.argsort(vrow, stable=True)
+    vs = vrow.index_select(0, order)
+    xs = vals.index_select(0, order).to(torch.float64)
+    cs = torch.zeros(len(xs) + 1, dtype=torch.float64, device=dev)
+    torch.cumsum(xs, 0, out=cs[1:])
+    ends = torch.searchsorted(
+        vs, torch.arange(nl + 1, dtype=vs.dtype, device=dev))
+    return cs.index_select(0, ends[1:]) - cs.index_select(0, ends[:-1])
+
+
 class _BaseHistGB(BaseEstimator):
     """NAMING NOTE: despite the ``HistGradientBoosting*`` class names
     (they ARE histogram-binned boosters), the parameter surface is
@@ -133,6 +170,23 @@ class _BaseHistGB(BaseEstimator):
         best_loss = np.inf
         stall = 0
 
+        # the boosting loop runs on DEVICE tensors end to end (round 1
+        # re-traversed X on the host twice per round — the measured fit
+        # was 75% host time, profiles/r02_boosting_profile.txt): raw
+        # scores F, gradients, leaf assignment (k_forest_apply against
+        # the resident X) and the Newton leaf sums all stay on ds.device
+        dev = ds.device
+        Ft = torch.as_tensor(F, dtype=torch.float64, device=dev)
+        self._to_device_targets(dev)
+        Xdev = None
+        if dev.type == "cuda":
+            Xdev = torch.as_tensor(
+                np.ascontiguousarray(X), device=dev)
+        val_t = (
+            torch.as_tensor(val_idx, dtype=torch.int64, device=dev)
+            if val_idx is not None else None
+        )
+
         stages = []
         for _ in range(n_rounds):
             if self.subsample < 1.0:
@@ -145,23 +199,27 @@ class _BaseHistGB(BaseEstimator):
                 mask = None
             if train_mask is not None:
                 mask = train_mask if mask is None else mask * train_mask
-            grad, hess = self._gradients(y, F)   # [n, K] each
+            mask_t = (
+                torch.as_tensor(mask, device=dev) if mask is not None
+                else None
+            )
+            grad, hess = self._gradients_t(Ft)   # [n, K] device fp64
             round_trees = []
             for k in range(K):
-                ds.y_f = torch.as_tensor(
-                    np.ascontiguousarray(grad[:, k], dtype=np.float32),
-                    device=ds.device,
-                )
+                ds.y_f = grad[:, k].to(torch.float32).contiguous()
                 seed = int(rng.randint(1 << 31))
                 tree = builder.build([seed], sample_weight=mask)[0]
-                self._newton_leaves(tree, X, grad[:, k], hess[:, k],
-                                    mask, K)
-                rows = _leaf_rows(tree, X)
-                F[:, k] += self.learning_rate * tree.value[rows, 0]
+                vrow = _leaf_value_rows(tree, X, Xdev, dev)
+                self._newton_leaves_t(tree, vrow, grad[:, k], hess[:, k],
+                                      mask_t, K)
+                val_col = torch.as_tensor(
+                    tree.value[:, 0], dtype=torch.float64, device=dev)
+                Ft[:, k] += self.learning_rate * val_col.index_select(
+                    0, vrow)
                 round_trees.append(tree)
             stages.append(round_trees)
-            if val_idx is not None:
-                loss = self._loss(val_idx, F)
+            if val_t is not None:
+                loss = self._loss_t(val_t, Ft)
                 if loss < best_loss - self.tol:
                     best_loss = loss
                     stall = 0
@@ -173,26 +231,28 @@ class _BaseHistGB(BaseEstimator):
         self.stages_ = stages
         self.n_features_in_ = X.shape[1]
         self.n_estimators_ = len(stages)
-        for a in ("_enc", "_yv"):  # training labels don't belong in the pickle
+        # training labels / device tensors don't belong in the pickle
+        for a in ("_enc", "_yv", "_enc_t", "_yv_t"):
             if hasattr(self, a):
                 delattr(self, a)
         _strip(self)
         return self
 
-    def _newton_leaves(self, tree, X, g, h, mask, K):
+    def _newton_leaves_t(self, tree, vrow, g, h, mask_t, K):
         """Refit leaf payloads with the one-step Newton update over the
-        (subsampled) rows: value = scale * Σg / Σh per leaf."""
+        (subsampled) rows: value = scale * Σg / Σh per leaf — computed
+        on device with a deterministic sort + cumsum segment sum."""
         scale = self._newton_scale(K)
         if scale is None:  # squared loss: mean residual is already right
             return
-        rows = _leaf_rows(tree, X)
-        w = mask if mask is not None else np.ones(len(g), dtype=np.float32)
+        if mask_t is not None:
+            g = g * mask_t
+            h = h * mask_t
         nl = len(tree.value)
-        num = np.bincount(rows, weights=g * w, minlength=nl)
-        den = np.bincount(rows, weights=h * w, minlength=nl)
-        tree.value = (
-            scale * num / np.clip(den, 1e-8, None)
-        ).astype(np.float32)[:, None]
+        num = _seg_sum(vrow, g, nl)
+        den = _seg_sum(vrow, h, nl)
+        val = scale * num / den.clamp_min(1e-8)
+        tree.value = val.to(torch.float32).cpu().numpy()[:, None]
 
     # ------------------------------------------------------------------ #
     def _raw_scores(self, X):
@@ -305,16 +365,20 @@ class HistGradientBoostingRegressor(RegressorMixin, _BaseHistGB):
         self.init_raw_ = float(self._yv.mean())
         return np.full((n, 1), self.init_raw_), 1
 
-    def _loss(self, idx, F):
-        d = self._yv[idx] - F[idx, 0]
-        return float(np.mean(d * d))
+    def _to_device_targets(self, dev):
+        self._yv_t = torch.as_tensor(self._yv, dtype=torch.float64,
+                                     device=dev)
+
+    def _loss_t(self, idx, Ft):
+        d = self._yv_t.index_select(0, idx) - Ft.index_select(0, idx)[:, 0]
+        return float((d * d).mean())
 
     def _base_raw(self):
         return np.array([self.init_raw_])
 
-    def _gradients(self, y, F):
-        g = (np.asarray(y, dtype=np.float64) - F[:, 0])[:, None]
-        return g, np.ones_like(g)
+    def _gradients_t(self, Ft):
+        g = (self._yv_t - Ft[:, 0]).unsqueeze(1)
+        return g, torch.ones_like(g)
 
     def _newton_scale(self, K):
         return None  # leaf mean of residuals IS the Newton step
@@ -351,31 +415,36 @@ class HistGradientBoostingClassifier(ClassifierMixin, _BaseHistGB):
     def _base_raw(self):
         return self.init_raw_
 
-    def _gradients(self, y, F):
-        if F.shape[1] == 1:
-            p = 1.0 / (1.0 + np.exp(-F[:, 0]))
-            g = (self._enc - p)[:, None]
-            h = (p * (1.0 - p))[:, None]
+    def _to_device_targets(self, dev):
+        self._enc_t = torch.as_tensor(self._enc, dtype=torch.int64,
+                                      device=dev)
+
+    def _gradients_t(self, Ft):
+        if Ft.shape[1] == 1:
+            p = torch.sigmoid(Ft[:, 0])
+            g = (self._enc_t.to(torch.float64) - p).unsqueeze(1)
+            h = (p * (1.0 - p)).unsqueeze(1)
             return g, h
-        e = np.exp(F - F.max(axis=1, keepdims=True))
-        P = e / e.sum(axis=1, keepdims=True)
-        Y = np.zeros_like(P)
-        Y[np.arange(len(self._enc)), self._enc] = 1.0
+        P = torch.softmax(Ft, dim=1)
+        Y = torch.zeros_like(P)
+        Y[torch.arange(len(self._enc_t), device=Ft.device),
+          self._enc_t] = 1.0
         return Y - P, P * (1.0 - P)
 
     def _newton_scale(self, K):
         return 1.0 if K == 1 else (K - 1.0) / K
 
-    def _loss(self, idx, F):
+    def _loss_t(self, idx, Ft):
         # mean deviance (neg log-likelihood) on the validation slice
-        if F.shape[1] == 1:
-            z = F[idx, 0]
-            t = self._enc[idx]
-            return float(np.mean(np.logaddexp(0.0, z) - t * z))
-        z = F[idx]
-        lse = np.log(np.exp(z - z.max(axis=1, keepdims=True)).sum(axis=1))
-        lse += z.max(axis=1)
-        return float(np.mean(lse - z[np.arange(len(idx)), self._enc[idx]]))
+        z = Ft.index_select(0, idx)
+        t = self._enc_t.index_select(0, idx)
+        if Ft.shape[1] == 1:
+            z0 = z[:, 0]
+            return float(
+                (torch.nn.functional.softplus(z0)
+                 - t.to(torch.float64) * z0).mean())
+        lse = torch.logsumexp(z, dim=1)
+        return float((lse - z.gather(1, t.unsqueeze(1))[:, 0]).mean())
 
     def decision_function(self, X):
         r = self._raw_scores(X)
