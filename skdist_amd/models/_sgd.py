@@ -46,6 +46,23 @@ LOSS_LOG, LOSS_HINGE, LOSS_SQUARED = 0, 1, 2
 _LOSS_IDS = {"log": LOSS_LOG, "hinge": LOSS_HINGE, "squared": LOSS_SQUARED}
 
 
+def encode_labels(y_np):
+    """(classes, int32 codes) — np.unique(return_inverse) semantics with
+    a bincount fast path for non-negative ints (np.unique sorts: ~7 ms
+    per 1M labels, several times per fit through sklearn's validators)."""
+    if y_np.dtype.kind in "iu" and y_np.size:
+        mn = int(y_np.min())
+        mx = int(y_np.max())
+        if 0 <= mn and mx < (1 << 22):
+            counts = np.bincount(y_np.ravel(), minlength=mx + 1)
+            cls = np.flatnonzero(counts)
+            lut = np.zeros(mx + 1, dtype=np.int32)
+            lut[cls] = np.arange(len(cls), dtype=np.int32)
+            return cls.astype(y_np.dtype), lut[y_np]
+    cls, enc = np.unique(y_np, return_inverse=True)
+    return cls, np.ascontiguousarray(enc, dtype=np.int32)
+
+
 def _use_hip(device):
     """HIP kernels are mandatory on GPU unless explicitly waived."""
     if device.type != "cuda":
@@ -130,7 +147,7 @@ class DeviceDataset:
                 y_host = np.ascontiguousarray(y_np, dtype=np.float32)
                 kind = "reg"
             elif task == "cls" or y_np.dtype.kind != "f":
-                cls_arr, enc = np.unique(y_np, return_inverse=True)
+                cls_arr, enc = encode_labels(y_np)
                 y_host = np.ascontiguousarray(enc, dtype=np.int32)
                 kind = "cls"
 

@@ -39,7 +39,7 @@ import warnings
 import numpy as np
 import torch
 
-from ._sgd import _LOSS_IDS, _MetricState, _use_hip
+from ._sgd import _LOSS_IDS, _MetricState, _use_hip, encode_labels
 
 LOSS_LOG, LOSS_HINGE, LOSS_SQUARED = 0, 1, 2
 
@@ -84,7 +84,7 @@ class SparseDeviceDataset:
                 y_host = np.ascontiguousarray(y_np, dtype=np.float32)
                 kind = "reg"
             else:
-                cls_arr, enc = np.unique(y_np, return_inverse=True)
+                cls_arr, enc = encode_labels(y_np)
                 y_host = np.ascontiguousarray(enc, dtype=np.int32)
                 kind = "cls"
 
