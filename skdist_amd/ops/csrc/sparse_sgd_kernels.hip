@@ -290,7 +290,7 @@ extern "C" hipError_t skdist_sp_sgd_step(
                            nch, cp, inv_m, lr_scale);
         HIP_CHECK(hipGetLastError());
     }
-    {
+    if (uf > 0) {
         const int tc = cp >= 256 ? 256 : cp;
         const int fpb = 256 / tc;
         dim3 grid((unsigned)((uf + fpb - 1) / fpb),
@@ -311,6 +311,7 @@ extern "C" hipError_t skdist_sp_forward(
     const void* W, const void* Wb, const void* s, const void* rows,
     void* Z, long long m, int cp, hipStream_t stream)
 {
+    if (m == 0) return hipSuccess;
     sp_fwd_launch(crow, cidx, cval, W, Wb, s, nullptr, nullptr, nullptr,
                   nullptr, nullptr, nullptr, rows, nullptr, Z, 0, m, cp,
                   0, stream);
