@@ -124,12 +124,12 @@ def main():
     t0 = time.time()
     cv_vec = CountVectorizer(max_features=30_000)
     Xc = cv_vec.fit_transform(docs_tr)
-    kb = SelectKBest(f_classif, k=min(2000, Xc.shape[1] - 1)).fit(
+    kb = SelectKBest(f_classif, k=min(1000, Xc.shape[1] - 1)).fit(
         Xc, y_tr)
     Xk = np.asarray(kb.transform(Xc).todense(), dtype=np.float32)
     ert = DistExtraTreesClassifier(
-        n_estimators=1000 if on_gpu else 50, random_state=0,
-        sc=sc_factory()).fit(Xk, y_tr)
+        n_estimators=1000 if on_gpu else 50, max_depth=20,
+        random_state=0, sc=sc_factory()).fit(Xk, y_tr)
 
     class _TreePipe:
         classes_ = ert.classes_

@@ -34,7 +34,7 @@ for t in m.estimators_:
     depths.append(mx)
 print(f"fit {fit_s:.1f}s; depths min/med/max: {min(depths)}/{int(np.median(depths))}/{max(depths)}")
 pred = DistPredictor(m, sc=Cluster(require_gpu=True), method="predict_proba")
-t0 = time.time(); p = pred.transform(X); dt = time.time() - t0
+t0 = time.time(); p = pred(X); dt = time.time() - t0
 print(f"deep-tree predict_proba 1M x 64 trees: {dt:.2f}s = {len(X)/dt/1e6:.1f}M rows/s")
 print("acc:", ((p[:, 1] > 0.5).astype(int) == y).mean())
 PYEOF
