@@ -101,16 +101,29 @@ class SparseDeviceDataset:
             self._shape = Xc.shape
 
         if cluster is not None and cluster.distributed:
+            import torch.distributed as dist
+
             dev = cluster.device
-            crow = cluster.bcast_tensor(
-                crow.to(dev) if crow is not None else None)
-            cidx = cluster.bcast_tensor(
-                cidx.to(dev) if cidx is not None else None)
-            cval = cluster.bcast_tensor(
-                cval.to(dev) if cval is not None else None)
-            yt = cluster.bcast_tensor(
-                torch.as_tensor(y_host).to(dev)
-                if y_host is not None else None)
+            # SPMD fast path (see DeviceDataset): all ranks holding the
+            # identical host CSR upload locally, skipping the broadcast
+            have = [None] * cluster.world_size
+            dist.all_gather_object(
+                have, crow is not None and y_host is not None)
+            if all(have):
+                crow = crow.to(dev)
+                cidx = cidx.to(dev)
+                cval = cval.to(dev)
+                yt = torch.as_tensor(y_host).to(dev)
+            else:
+                crow = cluster.bcast_tensor(
+                    crow.to(dev) if crow is not None else None)
+                cidx = cluster.bcast_tensor(
+                    cidx.to(dev) if cidx is not None else None)
+                cval = cluster.bcast_tensor(
+                    cval.to(dev) if cval is not None else None)
+                yt = cluster.bcast_tensor(
+                    torch.as_tensor(y_host).to(dev)
+                    if y_host is not None else None)
             kind, cls_arr, shape = cluster.bcast_obj(
                 (kind, cls_arr, getattr(self, "_shape", None)))
             if shape is not None:
