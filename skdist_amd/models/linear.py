@@ -651,6 +651,12 @@ class LogisticRegression(ClassifierMixin, _BatchedLinearBase):
     are mapped back to raw space, so the fitted model is exchangeable),
     and the optimizer is mini-batch SGD — scores match sklearn to CV-noise
     tolerance, not bitwise.
+
+    Wide/huge sparse X trains on the sparse-native solver
+    (models/_sparse_sgd.py), which by design skips standardization
+    (centering would densify; hashed text is already unit-scale), runs
+    momentum-free, and Adagrad-normalizes the data-gradient steps
+    unless ``adaptive=False``.
     """
 
     _loss = LOSS_LOG
