@@ -312,6 +312,17 @@ def sparse_sgd_fit(ds, spec, loss, epochs, batch_size, seed=0,
             sh["rw"] if sh["rw"] is not None
             else torch.empty(0, dtype=torch.float32, device=dev)
         )
+        # host-side fp32 mirror of the lazy-scale decay (same sequential
+        # multiplies as k_sp_bias_scale), so the renorm check never
+        # reads the device — epochs enqueue back-to-back with no syncs
+        s_host = np.ones(cp, dtype=np.float32)
+        lr_h = spec.col_lr.cpu().numpy().astype(np.float32)
+        l2_h = spec.col_l2.cpu().numpy().astype(np.float32)
+        lr_pad = np.zeros(cp, dtype=np.float32)
+        l2_pad = np.zeros(cp, dtype=np.float32)
+        lr_pad[: len(lr_h)] = lr_h
+        l2_pad[: len(l2_h)] = l2_h
+        n_batches = (ds.n + batch_size - 1) // batch_size
         for epoch in range(epochs):
             lr_scale = 1.0 / (1.0 + lr_decay * epoch)
             ext.sp_sgd_epoch(
@@ -321,10 +332,14 @@ def sparse_sgd_fit(ds, spec, loss, epochs, batch_size, seed=0,
                 col["l2"], sh["ufeat"], sh["cptr"], sh["ridx"],
                 sh["bval"], sh["ub_ptr"], sh["inv_m"],
                 int(batch_size), int(loss_id), float(lr_scale))
-            smin = float(s.min())
-            if smin < 1e-3:
+            f = (np.float32(1.0)
+                 - (lr_pad * np.float32(lr_scale)) * l2_pad)
+            for _ in range(n_batches):
+                s_host *= f
+            if float(s_host.min()) < 1e-3:
                 ext.sp_renorm(W[: ds.f], s)
                 s.fill_(1.0)
+                s_host[:] = 1.0
     else:
         _sparse_sgd_eager(ds, sh, W, Wb, s, H, Hb, col, loss_id, epochs,
                           batch_size, lr_decay, cp)

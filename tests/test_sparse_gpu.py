@@ -125,3 +125,22 @@ def test_text_scale_search_on_device():
     gs.fit(X, y)
     assert gs.best_score_ > 0.9, gs.best_score_
     assert gs.best_estimator_.coef_.shape == (1, 2 ** 20)
+
+
+def test_hip_heavy_l2_renorm_matches_eager():
+    """Heavy regularization drives the lazy scale through the renorm
+    path; HIP (host-mirrored scale, no syncs) must track the eager
+    mirror."""
+    ds, spec0, _, _ = _ds_and_spec(seed=4)
+    spec = ColumnSpec(
+        ds.device,
+        col_fold=spec0.col_fold.cpu().numpy(),
+        col_class=spec0.col_class.cpu().numpy(),
+        col_lr=np.full(spec0.ncols, 1.0, dtype=np.float32),
+        col_l2=np.full(spec0.ncols, 2e-3, dtype=np.float32),
+    )
+    W_hip = sparse_sgd_fit(ds, spec, "log", epochs=8, batch_size=512,
+                           seed=0).cpu().numpy()
+    W_eag = sparse_sgd_fit(ds, spec, "log", epochs=8, batch_size=512,
+                           seed=0, force_eager=True).cpu().numpy()
+    assert np.corrcoef(W_hip.ravel(), W_eag.ravel())[0, 1] > 0.999
