@@ -132,15 +132,17 @@ def test_hip_heavy_l2_renorm_matches_eager():
     path; HIP (host-mirrored scale, no syncs) must track the eager
     mirror."""
     ds, spec0, _, _ = _ds_and_spec(seed=4)
+    # decay 0.92/step x 100 steps -> s ~ 2e-4 < 1e-3: renorm fires
     spec = ColumnSpec(
         ds.device,
         col_fold=spec0.col_fold.cpu().numpy(),
         col_class=spec0.col_class.cpu().numpy(),
         col_lr=np.full(spec0.ncols, 1.0, dtype=np.float32),
-        col_l2=np.full(spec0.ncols, 2e-3, dtype=np.float32),
+        col_l2=np.full(spec0.ncols, 0.08, dtype=np.float32),
     )
-    W_hip = sparse_sgd_fit(ds, spec, "log", epochs=8, batch_size=512,
+    W_hip = sparse_sgd_fit(ds, spec, "log", epochs=10, batch_size=512,
                            seed=0).cpu().numpy()
-    W_eag = sparse_sgd_fit(ds, spec, "log", epochs=8, batch_size=512,
+    W_eag = sparse_sgd_fit(ds, spec, "log", epochs=10, batch_size=512,
                            seed=0, force_eager=True).cpu().numpy()
-    assert np.corrcoef(W_hip.ravel(), W_eag.ravel())[0, 1] > 0.999
+    assert np.isfinite(W_hip).all()
+    assert np.corrcoef(W_hip.ravel(), W_eag.ravel())[0, 1] > 0.99
