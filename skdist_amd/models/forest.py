@@ -1194,7 +1194,11 @@ def flat_forest_for(model, device):
         # flat forest — those models expose _device_predict_fn instead)
     trees = list(trees)
     if all(isinstance(t, HistTree) for t in trees):
-        return FlatForest(trees, device)
+        flat = FlatForest(trees, device)
+        # the traversal kernel carries <=32 payload values per leaf
+        # (predict_kernels.hip MAXVS); wider-class forests score on the
+        # host path instead of erroring at launch
+        return None if flat.vs > 32 else flat
     if all(hasattr(t, "tree_") for t in trees):
         is_cls = hasattr(trees[0], "predict_proba") and hasattr(
             model, "classes_")
@@ -1207,5 +1211,5 @@ def flat_forest_for(model, device):
             device,
         )
         flat.classes_ = getattr(model, "classes_", None)
-        return flat
+        return None if flat.vs > 32 else flat
     return None

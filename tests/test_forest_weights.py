@@ -126,3 +126,19 @@ def test_forest_sample_weight_cpu_path_still_exact():
     m = DistRandomForestClassifier(
         n_estimators=5, random_state=0).fit(X, y, sample_weight=sw)
     assert (m.predict(X) == y).mean() > 0.9
+
+
+def test_wide_class_forest_predicts_on_host():
+    """>32-class forests fall back to host scoring instead of raising
+    at the traversal kernel's MAXVS guard (latent round-1 crash)."""
+    from skdist_amd.distribute.ensemble import DistRandomForestClassifier
+    from skdist_amd.models.forest import flat_forest_for
+
+    rng = np.random.default_rng(0)
+    X = rng.standard_normal((1200, 8)).astype(np.float32)
+    y = rng.integers(0, 40, size=1200)
+    m = DistRandomForestClassifier(n_estimators=4, max_depth=6,
+                                   random_state=0).fit(X, y)
+    assert flat_forest_for(m, "cpu") is None  # 40-class payloads
+    p = m.predict_proba(X[:10])
+    assert p.shape == (10, 40)
