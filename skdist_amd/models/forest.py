@@ -1193,15 +1193,18 @@ def flat_forest_for(model, device):
         return None  # (a boosted model's [n_stages, K] array is not a
         # flat forest — those models expose _device_predict_fn instead)
     trees = list(trees)
+    # the traversal kernel carries <=32 payload values per leaf
+    # (predict_kernels.hip MAXVS); wider-class forests score on the
+    # host path instead of erroring at launch
     if all(isinstance(t, HistTree) for t in trees):
-        flat = FlatForest(trees, device)
-        # the traversal kernel carries <=32 payload values per leaf
-        # (predict_kernels.hip MAXVS); wider-class forests score on the
-        # host path instead of erroring at launch
-        return None if flat.vs > 32 else flat
+        if trees[0].value.shape[1] > 32:
+            return None
+        return FlatForest(trees, device)
     if all(hasattr(t, "tree_") for t in trees):
         is_cls = hasattr(trees[0], "predict_proba") and hasattr(
             model, "classes_")
+        if is_cls and len(model.classes_) > 32:
+            return None
         flat = FlatForest(
             [
                 _sklearn_tree_to_hist_tree(
@@ -1211,5 +1214,5 @@ def flat_forest_for(model, device):
             device,
         )
         flat.classes_ = getattr(model, "classes_", None)
-        return None if flat.vs > 32 else flat
+        return flat
     return None
