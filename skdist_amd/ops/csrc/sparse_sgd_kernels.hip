@@ -73,9 +73,13 @@ extern "C" __global__ __launch_bounds__(256) void k_sp_fwd(
 {
     const int tc = g_tc(cp);
     const int rpb = 256 / tc;
-    const long long ri = (long long)blockIdx.x * rpb + threadIdx.x / tc;
     const int c = blockIdx.y * tc + (threadIdx.x % tc);
-    if (ri >= m || c >= cp) return;
+    if (c >= cp) return;
+    // grid-stride over rows: short hashed-text rows make one-row-per-
+    // block launch/latency-bound (guide Guideline 11)
+    const long long stride = (long long)gridDim.x * rpb;
+    for (long long ri = (long long)blockIdx.x * rpb + threadIdx.x / tc;
+         ri < m; ri += stride) {
     const long long r = (rows != nullptr) ? rows[ri] : start + ri;
 
     const long long k0 = crow[r], k1 = crow[r + 1];
@@ -96,7 +100,7 @@ extern "C" __global__ __launch_bounds__(256) void k_sp_fwd(
 
     if (G == nullptr) {            // scoring path: raw decision values
         Z[ri * cp + c] = z;
-        return;
+        continue;
     }
     const float yv = y[r];
     const int cls = col_class[c];
@@ -108,6 +112,7 @@ extern "C" __global__ __launch_bounds__(256) void k_sp_fwd(
     if (c2 >= 0)
         train = train && (yv == (float)cls || yv == (float)c2);
     G[ri * cp + c] = f32_to_bf16(train ? g : 0.f);
+    }
 }
 
 // ---------------------------------------------------------------------- //
@@ -195,9 +200,13 @@ extern "C" __global__ __launch_bounds__(256) void k_sp_update(
 {
     const int tc = g_tc(cp);
     const int fpb = 256 / tc;
-    const long long si = (long long)blockIdx.x * fpb + threadIdx.x / tc;
     const int c = blockIdx.y * tc + (threadIdx.x % tc);
-    if (si >= uf || c >= cp) return;
+    if (c >= cp) return;
+    // grid-stride over feature slots: most hashed-text features have
+    // 1-4 batch rows, so one-slot-per-block is launch/latency-bound
+    const long long stride = (long long)gridDim.x * fpb;
+    for (long long si = (long long)blockIdx.x * fpb + threadIdx.x / tc;
+         si < uf; si += stride) {
     const int j = ufeat[si];
     const long long k0 = cptr[si], k1 = cptr[si + 1];
     float a0 = 0.f, a1 = 0.f, a2 = 0.f, a3 = 0.f;
@@ -219,6 +228,7 @@ extern "C" __global__ __launch_bounds__(256) void k_sp_update(
         step *= __frsqrt_rn(hv + 1e-12f);
     }
     W[e] -= step / s[c];
+    }
 }
 
 // fold the scale into W when it drifts far from 1 (rare; keeps /s[c]
@@ -244,8 +254,9 @@ static inline void sp_fwd_launch(
 {
     const int tc = cp >= 256 ? 256 : cp;
     const int rpb = 256 / tc;
-    dim3 grid((unsigned)((m + rpb - 1) / rpb),
-              (unsigned)((cp + tc - 1) / tc));
+    long long gx = (m + rpb - 1) / rpb;
+    if (gx > 2048) gx = 2048;
+    dim3 grid((unsigned)gx, (unsigned)((cp + tc - 1) / tc));
     hipLaunchKernelGGL(k_sp_fwd, grid, dim3(256), 0, st,
                        (const long long*)crow, (const int*)cidx,
                        (const float*)cval, (const float*)W,
@@ -293,8 +304,9 @@ extern "C" hipError_t skdist_sp_sgd_step(
     if (uf > 0) {
         const int tc = cp >= 256 ? 256 : cp;
         const int fpb = 256 / tc;
-        dim3 grid((unsigned)((uf + fpb - 1) / fpb),
-                  (unsigned)((cp + tc - 1) / tc));
+        long long gx = (uf + fpb - 1) / fpb;
+        if (gx > 2048) gx = 2048;
+        dim3 grid((unsigned)gx, (unsigned)((cp + tc - 1) / tc));
         hipLaunchKernelGGL(k_sp_update, grid, dim3(256), 0, stream,
                            (const int*)ufeat, (const long long*)cptr,
                            (const int*)ridx, (const float*)bval,
