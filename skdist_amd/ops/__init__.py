@@ -192,12 +192,22 @@ def hip_sgd_solve(ds, spec, loss_id, epochs, batch_size, seed, momentum,
         ]
         inv_m = torch.tensor(
             [1.0 / max(v, 1e-30) for v in sums], dtype=torch.float32)
-    for epoch in range(epochs):
-        lr_scale = 1.0 / (1.0 + lr_decay * epoch)
-        ext.sgd_epoch(
+    if lr_decay == 0.0:
+        # constant lr_scale: every epoch launches the identical
+        # sequence — one C++ call, epochs 2..N replay a hipGraph
+        ext.sgd_solve(
             Xs, XsT, GT, W, V, WbfT, partial, ys, folds,
             cls_p, cfold_p, cls2_p, lr_p, l2_p, fmask, rw_t, inv_m,
-            bs, int(loss_id), float(lr_scale), float(momentum),
-            int(ds.intercept_row),
+            bs, int(loss_id), float(momentum), int(ds.intercept_row),
+            int(epochs),
         )
+    else:
+        for epoch in range(epochs):
+            lr_scale = 1.0 / (1.0 + lr_decay * epoch)
+            ext.sgd_epoch(
+                Xs, XsT, GT, W, V, WbfT, partial, ys, folds,
+                cls_p, cfold_p, cls2_p, lr_p, l2_p, fmask, rw_t, inv_m,
+                bs, int(loss_id), float(lr_scale), float(momentum),
+                int(ds.intercept_row),
+            )
     return W[:, :ncols]
