@@ -110,7 +110,16 @@ static void run_epochs_graphed(int64_t epochs, hipStream_t torch_stream,
         bool ok = hipStreamBeginCapture(
                       s2, hipStreamCaptureModeThreadLocal) == hipSuccess;
         if (ok) {
-            body(s2);  // recorded, not executed
+            try {
+                body(s2);  // recorded, not executed
+            } catch (...) {
+                // never leave the stream in capture mode
+                hipStreamEndCapture(s2, &graph);
+                if (graph) hipGraphDestroy(graph);
+                hipEventDestroy(ev);
+                hipStreamDestroy(s2);
+                throw;
+            }
             ok = hipStreamEndCapture(s2, &graph) == hipSuccess;
         }
         if (ok)
