@@ -311,13 +311,58 @@ class DistOneVsRestClassifier(ClassifierMixin, BaseEstimator):
         dev = torch.device(device)
         Wt = torch.as_tensor(W.T.copy(), device=dev)   # [f, k]
         bt = torch.as_tensor(b, device=dev)
+        sp_state = {}
+
+        def _scores_sparse(Xc):
+            # hashed-text OvR serving: CSR chunks through the sparse
+            # forward kernel (k columns padded to 64)
+            from ..ops import require_hip
+
+            if not sp_state:
+                k = Wt.shape[1]
+                kp = (k + 63) // 64 * 64
+                Wsp = torch.zeros(Wt.shape[0], kp, dtype=torch.float32,
+                                  device=dev)
+                Wsp[:, :k] = Wt.to(torch.float32)
+                bsp = torch.zeros(kp, dtype=torch.float32, device=dev)
+                bsp[:k] = bt.to(torch.float32)
+                sp_state.update(
+                    W=Wsp.contiguous(), b=bsp.contiguous(),
+                    s=torch.ones(kp, dtype=torch.float32, device=dev),
+                    k=k)
+            crow = torch.as_tensor(
+                np.ascontiguousarray(Xc.indptr, dtype=np.int64),
+                device=dev)
+            cidx = torch.as_tensor(
+                np.ascontiguousarray(Xc.indices, dtype=np.int32),
+                device=dev)
+            cval = torch.as_tensor(
+                np.ascontiguousarray(Xc.data, dtype=np.float32),
+                device=dev)
+            rows = torch.arange(Xc.shape[0], dtype=torch.int64,
+                                device=dev)
+            Z = torch.empty(Xc.shape[0], sp_state["W"].shape[1],
+                            dtype=torch.float32, device=dev)
+            require_hip().sp_forward(crow, cidx, cval, sp_state["W"],
+                                     sp_state["b"], sp_state["s"],
+                                     rows, Z)
+            return Z[:, : sp_state["k"]]
 
         def fn(X, chunk=1 << 21):
-            X = np.ascontiguousarray(X, dtype=np.float32)
+            import scipy.sparse as _sp
+
+            is_sp = _sp.issparse(X)
+            if is_sp:
+                X = X.tocsr()
+            else:
+                X = np.ascontiguousarray(X, dtype=np.float32)
             outs = []
-            for lo in range(0, len(X), chunk):
-                xb = torch.as_tensor(X[lo: lo + chunk], device=dev)
-                Z = xb @ Wt + bt
+            for lo in range(0, X.shape[0], chunk):
+                if is_sp:
+                    Z = _scores_sparse(X[lo: lo + chunk])
+                else:
+                    xb = torch.as_tensor(X[lo: lo + chunk], device=dev)
+                    Z = xb @ Wt + bt
                 if has_proba:
                     Z = torch.sigmoid(Z)
                 if method == "predict":

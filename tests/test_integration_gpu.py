@@ -75,3 +75,31 @@ def test_multimodel_search_native_models_on_device():
     ms.fit(X, y)
     assert ms.best_estimator_ is not None
     assert (ms.predict(X[:200]) == y[:200]).mean() > 0.7
+
+
+def test_ovr_text_serving_sparse_device():
+    """Fitted OvR over hashed text served through DistPredictor with
+    CSR input: the device hook's sparse branch must match host scoring."""
+    import scipy.sparse as sp
+
+    from skdist_amd.distribute.predict import DistPredictor
+
+    rng = np.random.default_rng(3)
+    n, f, k = 30_000, 1 << 17, 4
+    rows = np.repeat(np.arange(n), 15)
+    cols = np.sort(rng.integers(0, f, size=(n, 15)), axis=1).ravel()
+    vals = np.full(n * 15, 0.26, dtype=np.float32)
+    X = sp.csr_matrix((vals, (rows, cols.astype(np.int64))),
+                      shape=(n, f))
+    W = np.zeros((f, k), dtype=np.float32)
+    W[: 4096] = rng.standard_normal((4096, k))
+    y = np.asarray(X @ W).argmax(axis=1)
+    ovr = DistOneVsRestClassifier(
+        LogisticRegression(epochs=8, momentum=0.0, random_state=0),
+        sc=Cluster(require_gpu=True)).fit(X, y)
+    pred = DistPredictor(ovr, sc=Cluster(require_gpu=True),
+                         method="predict")
+    p_dev = pred(X[:4000])
+    p_host_fn = getattr(ovr, "predict")
+    agree = (p_dev == p_host_fn(X[:4000])).mean()
+    assert agree > 0.999, agree
