@@ -47,6 +47,36 @@ from .validation import _check_estimator
 import warnings
 
 
+def _slice_fit_params(fit_params, indices, n_samples):
+    """Per-fold view of fit params: any array-like whose first dimension
+    matches ``n_samples`` (sample_weight etc.) is indexed by the fold's
+    training rows, everything else passes through — sklearn's
+    ``_check_method_params`` routing semantics (the reference relied on
+    Spark shipping the full closure and had the same latent mismatch)."""
+    if not fit_params:
+        return fit_params
+    import scipy.sparse as _sp
+
+    out = {}
+    idx = np.asarray(indices)
+    for key, val in fit_params.items():
+        try:
+            if _sp.issparse(val) and val.shape[0] == n_samples:
+                out[key] = val[idx]
+                continue
+            if hasattr(val, "iloc") and len(val) == n_samples:
+                out[key] = val.iloc[idx]
+                continue
+            if hasattr(val, "__len__") and not isinstance(val, str) \
+                    and len(val) == n_samples:
+                out[key] = np.asarray(val)[idx]
+                continue
+        except TypeError:
+            pass
+        out[key] = val
+    return out
+
+
 def _fit_and_score_task(base_estimator, X, y, scorers, task, error_score,
                         return_train_score, fit_params, verbose=0):
     """Fit one (params, fold) task and score it — the generic worker
@@ -66,14 +96,15 @@ def _fit_and_score_task(base_estimator, X, y, scorers, task, error_score,
 
     X_train, y_train = _safe_split(est, X, y, train_idx)
     X_test, y_test = _safe_split(est, X, y, test_idx, train_idx)
+    fold_params = _slice_fit_params(fit_params, train_idx, _num_samples(X))
 
     start = time.perf_counter()
     result = {"task_id": task_id, "n_test": _num_samples(X_test)}
     try:
         if y_train is None:
-            est.fit(X_train, **fit_params)
+            est.fit(X_train, **fold_params)
         else:
-            est.fit(X_train, y_train, **fit_params)
+            est.fit(X_train, y_train, **fold_params)
     except Exception as e:
         result["fit_time"] = time.perf_counter() - start
         if error_score == "raise":
@@ -94,10 +125,18 @@ def _fit_and_score_task(base_estimator, X, y, scorers, task, error_score,
     result["fit_time"] = time.perf_counter() - start
 
     start = time.perf_counter()
-    result["test_scores"] = _score(est, X_test, y_test, scorers)
+    sw = fit_params.get("sample_weight") if fit_params else None
+    n_all = _num_samples(X)
+    result["test_scores"] = _score(
+        est, X_test, y_test, scorers,
+        sample_weight=_slice_fit_params(
+            {"sample_weight": sw}, test_idx, n_all)["sample_weight"]
+        if sw is not None else None)
     result["score_time"] = time.perf_counter() - start
     if return_train_score:
-        result["train_scores"] = _score(est, X_train, y_train, scorers)
+        result["train_scores"] = _score(
+            est, X_train, y_train, scorers,
+            sample_weight=fold_params.get("sample_weight"))
     return result
 
 
@@ -280,7 +319,8 @@ class DistBaseSearchCV(BaseEstimator, MetaEstimatorMixin, metaclass=ABCMeta):
             est.set_params(**self.best_params_)
             X_tr, y_tr = _safe_split(est, X, y, train_idx)
             X_te, _ = _safe_split(est, X, y, test_idx, train_idx)
-            est.fit(X_tr, y_tr, **fit_params)
+            est.fit(X_tr, y_tr, **_slice_fit_params(
+                fit_params, train_idx, _num_samples(X)))
             if hasattr(est, "predict_proba"):
                 p = est.predict_proba(X_te)
             else:

@@ -42,18 +42,40 @@ def _check_multimetric_scoring(estimator, scoring=None):
     raise ValueError(f"invalid scoring: {scoring!r}")
 
 
-def _score(estimator, X_test, y_test, scorers):
+def _scorer_accepts_sample_weight(scorer):
+    """sklearn 1.7 semantics: a Scorer advertises acceptance via
+    _accept_sample_weight(); bare callables by signature."""
+    import inspect
+
+    if hasattr(scorer, "_accept_sample_weight"):
+        try:
+            return bool(scorer._accept_sample_weight())
+        except Exception:
+            return False
+    try:
+        return "sample_weight" in inspect.signature(scorer).parameters
+    except (TypeError, ValueError):
+        return False
+
+
+def _score(estimator, X_test, y_test, scorers, sample_weight=None):
     """Score a fitted estimator on a test fold with one or more scorers.
 
     Returns {name: float}.  Mirrors reference utils.py:45-72 including the
-    "score must be a number" check.
+    "score must be a number" check.  ``sample_weight`` (already sliced to
+    the fold) forwards to each scorer that accepts it — sklearn's
+    routing-disabled BaseSearchCV behavior.
     """
     out = {}
     for name, scorer in scorers.items():
+        kw = {}
+        if sample_weight is not None and _scorer_accepts_sample_weight(
+                scorer):
+            kw["sample_weight"] = sample_weight
         if y_test is None:
-            s = scorer(estimator, X_test)
+            s = scorer(estimator, X_test, **kw)
         else:
-            s = scorer(estimator, X_test, y_test)
+            s = scorer(estimator, X_test, y_test, **kw)
         if hasattr(s, "item"):
             s = s.item()
         if not isinstance(s, numbers.Number):

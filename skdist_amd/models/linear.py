@@ -195,6 +195,15 @@ class _BatchedLinearBase(BaseEstimator):
         """
         if return_train_score:
             raise FallbackToGeneric("return_train_score not batched yet")
+        if sample_weight is not None:
+            # sklearn forwards an explicit fit-param sample_weight to the
+            # SCORERS too (test-sliced); the device scorers are
+            # unweighted, so the generic path keeps exact parity.
+            # (class_weight stays batched — sklearn never weights
+            # scoring for it.)
+            raise FallbackToGeneric(
+                "sample_weight weights the test scores in sklearn: "
+                "generic path")
         metric = self._device_metric(scoring)
         # single comprehension: the set().union(*(set(p) ...)) form
         # measured 40 ms/fit at 500 candidates (profiles/r4 host trace)

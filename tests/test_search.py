@@ -177,9 +177,11 @@ def test_out_of_fold_preds(small_xy):
 
 
 def test_search_with_sample_weight_batched(small_xy):
-    """fit_params={'sample_weight': w} rides the batched path: weights
-    reach the fused gradient epilogue instead of forcing per-task
-    fallback."""
+    """fit_params={'sample_weight': w} through the search.  As of
+    round 2 an explicit sample_weight routes to the GENERIC path
+    (sklearn forwards it to the scorers too, test-sliced — the device
+    scorers are unweighted); per-task fits still use the fused weight
+    plane via est.fit(sample_weight=...)."""
     from skdist_amd import Cluster
     from skdist_amd.models import LogisticRegression
 
@@ -202,10 +204,32 @@ def test_search_with_sample_weight_batched(small_xy):
         LogisticRegression(epochs=10, random_state=0),
         {"C": [1.0]}, cv=3, sc=Cluster())
     g2.fit(X, yb, sample_weight=w0)
-    # scoring counts all rows (sklearn scorers are unweighted), but the
-    # fitted models must be identical: zero-weight rows have no gradient
+    # the fitted models must be identical: zero-weight rows have no
+    # gradient (and with weighted scoring they don't count there either)
     np.testing.assert_allclose(
         g1.best_estimator_.coef_, g2.best_estimator_.coef_, atol=1e-7)
+
+
+def test_search_sample_weight_matches_sklearn_exactly():
+    """Generic-path fit AND scoring with sample_weight reproduce
+    sklearn's GridSearchCV bit-for-bit (round-2 fix: fold-sliced fit
+    params + weighted scorer forwarding; previously a shape crash)."""
+    from sklearn.linear_model import LogisticRegression as SkLR
+    from sklearn.model_selection import GridSearchCV
+
+    rng = np.random.default_rng(0)
+    X = rng.standard_normal((300, 5))
+    y = (X[:, 0] > 0).astype(int)
+    w = rng.random(300)
+    grid = {"C": [0.1, 1.0]}
+    ours = DistGridSearchCV(
+        SkLR(max_iter=100), grid, cv=3).fit(X, y, sample_weight=w)
+    theirs = GridSearchCV(
+        SkLR(max_iter=100), grid, cv=3).fit(X, y, sample_weight=w)
+    np.testing.assert_allclose(
+        ours.cv_results_["mean_test_score"],
+        theirs.cv_results_["mean_test_score"], atol=1e-14)
+    assert ours.best_index_ == theirs.best_index_
 
 
 def test_fit_params_passthrough_host_estimator():
