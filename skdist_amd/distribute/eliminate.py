@@ -134,7 +134,11 @@ class DistFeatureEliminator(ClassifierMixin, BaseEstimator):
                 Xs = _drop_cols(X, removals[si])
                 X_tr, y_tr = _safe_split(est, Xs, y, train)
                 X_te, y_te = _safe_split(est, Xs, y, test, train)
-                est.fit(X_tr, y_tr, **fit_params)
+                from .search import _slice_fit_params
+                from .utils import _num_samples
+
+                est.fit(X_tr, y_tr, **_slice_fit_params(
+                    fit_params, train, _num_samples(Xs)))
                 return tid, si, float(scorer(est, X_te, y_te))
 
             if sc is None:

@@ -565,7 +565,8 @@ class DistMultiModelSearch(BaseEstimator, MetaEstimatorMixin):
                 est.set_params(**params)
             X_tr, y_tr = _safe_split(est, X, y, train)
             X_te, y_te = _safe_split(est, X, y, test, train)
-            est.fit(X_tr, y_tr, **fit_params)
+            est.fit(X_tr, y_tr, **_slice_fit_params(
+                fit_params, train, _num_samples(X)))
             from sklearn.metrics import check_scoring
 
             scorer = check_scoring(est, scoring=self.scoring)

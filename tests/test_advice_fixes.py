@@ -155,3 +155,41 @@ def test_run_tasks_wraps_failures_in_world1():
     with pytest.raises(TaskFailedError, match="kaboom"):
         sc.run_tasks(boom, [0, 1, 2])
     assert sc.run_tasks(lambda t: t + 1, [1, 2, 3]) == [2, 3, 4]
+
+
+def test_fit_param_slicing_all_meta_estimators():
+    """Round-2 audit fix: sample-aligned fit params (sample_weight) are
+    fold-sliced in EVERY per-task worker (search / multimodel /
+    eliminator) — previously a shape crash."""
+    from sklearn.linear_model import LogisticRegression as SkLR
+
+    from skdist_amd.distribute.eliminate import DistFeatureEliminator
+    from skdist_amd.distribute.search import (
+        DistGridSearchCV,
+        DistMultiModelSearch,
+    )
+
+    rng = np.random.default_rng(0)
+    X = rng.standard_normal((240, 5))
+    y = (X[:, 0] > 0).astype(int)
+    w = rng.random(240)
+    gs = DistGridSearchCV(SkLR(max_iter=100), {"C": [1.0]}, cv=3)
+    gs.fit(X, y, sample_weight=w)
+    assert gs.best_score_ > 0.8
+    mm = DistMultiModelSearch(
+        [("lr", SkLR(max_iter=100), {"C": [0.1, 1.0]})], n=2, cv=3)
+    mm.fit(X, y, sample_weight=w)
+    assert mm.best_score_ > 0.8
+    fe = DistFeatureEliminator(
+        SkLR(max_iter=100), min_features_to_select=3, cv=3)
+    fe.fit(X, y, sample_weight=w)
+    assert fe.best_score_ > 0.8
+    # non-sample-aligned fit params pass through untouched
+    from skdist_amd.distribute.search import _slice_fit_params
+
+    fp = {"eval_set": [(X[:10], y[:10])], "classes": [0, 1],
+          "sample_weight": w}
+    out = _slice_fit_params(fp, np.arange(100), 240)
+    assert out["eval_set"] is fp["eval_set"]
+    assert out["classes"] is fp["classes"]
+    assert len(out["sample_weight"]) == 100
