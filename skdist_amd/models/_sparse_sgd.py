@@ -134,6 +134,10 @@ class SparseDeviceDataset:
             cval = cval.to(self.device)
             yt = torch.as_tensor(y_host).to(self.device)
 
+        from ._sgd import check_finite
+
+        check_finite(cval, "X (sparse values)")
+        check_finite(yt, "y")
         self.crow, self.cidx, self.cval = crow, cidx, cval
         self.n = crow.shape[0] - 1
         self.f = (

@@ -156,6 +156,9 @@ class BinnedDataset:
         self.n, self.f = X.shape
         self.nbins = nbins
         Xt = torch.as_tensor(X, device=self.device)
+        from ._sgd import check_finite
+
+        check_finite(Xt, "X")
 
         # per-feature quantile edges from a (seeded) row subsample
         if self.n > max_bin_sample:

@@ -193,3 +193,38 @@ def test_fit_param_slicing_all_meta_estimators():
     assert out["eval_set"] is fp["eval_set"]
     assert out["classes"] is fp["classes"]
     assert len(out["sample_weight"]) == 100
+
+
+def test_nan_inputs_raise_like_sklearn():
+    """Round-2 audit fix: NaN/inf inputs raise instead of silently
+    training NaN models (device-side finite check in every dataset
+    builder)."""
+    import scipy.sparse as sp
+
+    from skdist_amd.models import (
+        HistGradientBoostingClassifier,
+        LogisticRegression,
+        Ridge,
+    )
+
+    rng = np.random.default_rng(0)
+    X = rng.standard_normal((100, 4)).astype(np.float32)
+    y = (X[:, 0] > 0).astype(int)
+    Xn = X.copy()
+    Xn[3, 2] = np.nan
+    with pytest.raises(ValueError, match="NaN or infinity"):
+        LogisticRegression(epochs=2).fit(Xn, y)
+    with pytest.raises(ValueError, match="NaN or infinity"):
+        Ridge(epochs=2).fit(
+            X, np.where(np.arange(100) == 5, np.inf, y.astype(float)))
+    with pytest.raises(ValueError, match="NaN or infinity"):
+        HistGradientBoostingClassifier(n_estimators=2).fit(Xn, y)
+    with pytest.raises(ValueError, match="NaN or infinity"):
+        import os
+
+        os.environ["SKDIST_AMD_FORCE_SPARSE"] = "1"
+        try:
+            LogisticRegression(epochs=2, momentum=0).fit(
+                sp.csr_matrix(Xn), y)
+        finally:
+            os.environ.pop("SKDIST_AMD_FORCE_SPARSE")
