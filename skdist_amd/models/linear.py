@@ -113,6 +113,12 @@ class _BatchedLinearBase(BaseEstimator):
         ds.set_cv_partition([])  # no CV mask
         if is_clf:
             k = len(ds.classes_)
+            if k < 2:
+                raise ValueError(
+                    "This solver needs samples of at least 2 classes in "
+                    f"the data, but the data contains only one class: "
+                    f"{ds.classes_[0]!r}"
+                )
             ncols = 1 if k == 2 else k
             col_class = (
                 np.array([1], dtype=np.int32) if k == 2
@@ -236,6 +242,24 @@ class _BatchedLinearBase(BaseEstimator):
         n_cand = len(candidate_params)
         fold_train_n = [len(tr) for tr, _ in cv_splits]
         fold_test_n = [len(te) for _, te in cv_splits]
+
+        if is_clf and ds.classes_ is not None and len(ds.classes_) >= 2:
+            # a training fold missing one of the global classes makes
+            # sklearn raise in that task (error_score semantics); the
+            # joint batched solve would instead silently train a
+            # degenerate column — defer to the exact generic path.
+            import torch as _t
+
+            k_all = len(ds.classes_)
+            flat = ds.fold_id.to(_t.int64) * k_all + ds.y_int.to(_t.int64)
+            fold_counts = _t.bincount(
+                flat, minlength=n_folds * k_all
+            ).reshape(n_folds, k_all)
+            train_counts = fold_counts.sum(dim=0, keepdim=True) - fold_counts
+            if bool((train_counts == 0).any()):
+                raise FallbackToGeneric(
+                    "a training fold is missing a class: generic path "
+                    "(sklearn error_score semantics)")
 
         if is_clf:
             k = len(ds.classes_)
