@@ -189,6 +189,9 @@ class DistFeatureEliminator(ClassifierMixin, BaseEstimator):
         return X
 
     def predict(self, X):
+        from .validation import _require_fitted
+
+        _require_fitted(self, "best_estimator_")
         return self.best_estimator_.predict(self._mask_X(X))
 
     def predict_proba(self, X):

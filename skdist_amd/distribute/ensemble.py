@@ -407,6 +407,9 @@ class _ForestClassifierMixin(ClassifierMixin):
     _is_classifier = True
 
     def predict_proba(self, X):
+        from .validation import _require_fitted
+
+        _require_fitted(self, "estimators_")
         flat = self._device_forest()
         if flat is not None:
             return flat.predict_proba(X)
@@ -419,6 +422,9 @@ class _ForestClassifierMixin(ClassifierMixin):
         return proba / len(self.estimators_)
 
     def predict(self, X):
+        from .validation import _require_fitted
+
+        _require_fitted(self, "estimators_")
         return self.classes_[self.predict_proba(X).argmax(axis=1)]
 
     def predict_log_proba(self, X):
@@ -429,6 +435,9 @@ class _ForestRegressorMixin(RegressorMixin):
     _is_classifier = False
 
     def predict(self, X):
+        from .validation import _require_fitted
+
+        _require_fitted(self, "estimators_")
         flat = self._device_forest()
         if flat is not None:
             return flat.predict_value(X)[:, 0]

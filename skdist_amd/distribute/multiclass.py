@@ -227,6 +227,9 @@ class DistOneVsRestClassifier(ClassifierMixin, BaseEstimator):
         return None if fn is None else fn(X)
 
     def predict(self, X):
+        from .validation import _require_fitted
+
+        _require_fitted(self, "estimators_")
         if self.mlb_ is None and not getattr(self, "multilabel_", False):
             out = self._try_device("predict", X)
             if out is not None:
@@ -254,6 +257,9 @@ class DistOneVsRestClassifier(ClassifierMixin, BaseEstimator):
         """Vectorized replacement for reference multiclass.py:337-362;
         on a GPU machine the k binary columns come from one device GEMM
         with the sigmoid+normalize fused."""
+        from .validation import _require_fitted
+
+        _require_fitted(self, "estimators_")
         out = self._try_device("predict_proba", X)
         if out is not None:
             return out
@@ -453,6 +459,9 @@ class DistOneVsOneClassifier(ClassifierMixin, BaseEstimator):
     def predict(self, X):
         import torch
 
+        from .validation import _require_fitted
+
+        _require_fitted(self, "estimators_")
         if not sp.issparse(X) and torch.cuda.is_available():
             fn = self._device_predict_fn("predict", "cuda")
             if fn is not None:

@@ -404,23 +404,29 @@ class DistBaseSearchCV(BaseEstimator, MetaEstimatorMixin, metaclass=ABCMeta):
     def classes_(self):
         return self.best_estimator_.classes_
 
+    def _best(self):
+        from .validation import _require_fitted
+
+        _require_fitted(self, "best_estimator_")
+        return self.best_estimator_
+
     def predict(self, X):
-        return self.best_estimator_.predict(X)
+        return self._best().predict(X)
 
     def predict_proba(self, X):
-        return self.best_estimator_.predict_proba(X)
+        return self._best().predict_proba(X)
 
     def predict_log_proba(self, X):
-        return self.best_estimator_.predict_log_proba(X)
+        return self._best().predict_log_proba(X)
 
     def decision_function(self, X):
-        return self.best_estimator_.decision_function(X)
+        return self._best().decision_function(X)
 
     def transform(self, X):
-        return self.best_estimator_.transform(X)
+        return self._best().transform(X)
 
     def inverse_transform(self, X):
-        return self.best_estimator_.inverse_transform(X)
+        return self._best().inverse_transform(X)
 
     def score(self, X, y=None):
         if self.scorer_ is None:

@@ -69,3 +69,17 @@ def _check_n_iter(n_iter, param_distributions):
         size = int(np.prod([len(v) for v in param_distributions.values()]))
         return min(n_iter, size)
     return n_iter
+
+
+def _require_fitted(est, attr):
+    """Raise sklearn's NotFittedError when ``attr`` is absent — our
+    re-implemented predict paths previously surfaced a bare
+    AttributeError (NotFittedError subclasses it, so ``except
+    NotFittedError`` in user code would not fire)."""
+    if not hasattr(est, attr):
+        from sklearn.exceptions import NotFittedError
+
+        raise NotFittedError(
+            f"This {type(est).__name__} instance is not fitted yet. "
+            "Call 'fit' before using this estimator."
+        )

@@ -258,6 +258,13 @@ class _BaseHistGB(BaseEstimator):
     def _raw_scores(self, X):
         import scipy.sparse as sp
 
+        if not hasattr(self, "stages_"):
+            from sklearn.exceptions import NotFittedError
+
+            raise NotFittedError(
+                f"This {type(self).__name__} instance is not fitted yet. "
+                "Call 'fit' before using this estimator.")
+
         if sp.issparse(X):
             X = X.toarray()
         X = np.asarray(X, dtype=np.float32)

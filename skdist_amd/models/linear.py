@@ -173,6 +173,13 @@ class _BatchedLinearBase(BaseEstimator):
     def decision_function(self, X):
         import scipy.sparse as _sp
 
+        if not hasattr(self, "coef_"):
+            from sklearn.exceptions import NotFittedError
+
+            raise NotFittedError(
+                f"This {type(self).__name__} instance is not fitted yet. "
+                "Call 'fit' before using this estimator.")
+
         if _sp.issparse(X):
             scores = (
                 np.asarray(X.astype(np.float64) @ self.coef_.T)

@@ -228,3 +228,32 @@ def test_nan_inputs_raise_like_sklearn():
                 sp.csr_matrix(Xn), y)
         finally:
             os.environ.pop("SKDIST_AMD_FORCE_SPARSE")
+
+
+def test_predict_before_fit_raises_notfitted():
+    """Round-2 audit fix: unfitted predict raises sklearn's
+    NotFittedError (was a bare AttributeError, invisible to
+    ``except NotFittedError``)."""
+    from sklearn.exceptions import NotFittedError
+
+    from skdist_amd.distribute.eliminate import DistFeatureEliminator
+    from skdist_amd.distribute.ensemble import DistRandomForestClassifier
+    from skdist_amd.distribute.multiclass import (
+        DistOneVsOneClassifier,
+        DistOneVsRestClassifier,
+    )
+    from skdist_amd.distribute.search import DistGridSearchCV
+    from skdist_amd.models import (
+        HistGradientBoostingClassifier,
+        LogisticRegression,
+    )
+
+    X = np.zeros((4, 3), dtype=np.float32)
+    for m in (LogisticRegression(), HistGradientBoostingClassifier(),
+              DistRandomForestClassifier(n_estimators=2),
+              DistOneVsRestClassifier(LogisticRegression()),
+              DistOneVsOneClassifier(LogisticRegression()),
+              DistFeatureEliminator(LogisticRegression()),
+              DistGridSearchCV(LogisticRegression(), {"C": [1.0]})):
+        with pytest.raises(NotFittedError):
+            m.predict(X)
