@@ -14,6 +14,19 @@ import numpy as np
 from skdist_amd.distribute.eliminate import DistFeatureEliminator
 from skdist_amd.models import LogisticRegression
 
+
+def _sc():
+    """Cluster() on a GPU node, None for the local CPU path."""
+    import torch
+
+    if not torch.cuda.is_available():
+        return None
+    from skdist_amd import Cluster
+
+    return Cluster()
+
+
+
 rng = np.random.default_rng(0)
 n, f = 20000, 40
 X = rng.standard_normal((n, f)).astype(np.float32)
@@ -24,7 +37,7 @@ X[:, 12:] = rng.standard_normal((n, f - 12))
 
 el = DistFeatureEliminator(
     LogisticRegression(epochs=15, random_state=0),
-    sc=None,  # Cluster() on a GPU node -> one masked batched solve
+    sc=_sc(),  # Cluster() on a GPU node -> one masked batched solve
     min_features_to_select=8, step=4, cv=5)
 el.fit(X, y)
 print("kept features:", el.best_features_)

@@ -20,15 +20,28 @@ from skdist_amd.distribute.multiclass import (
 )
 from skdist_amd.models import LogisticRegression
 
+
+def _sc():
+    """Cluster() on a GPU node, None for the local CPU path."""
+    import torch
+
+    if not torch.cuda.is_available():
+        return None
+    from skdist_amd import Cluster
+
+    return Cluster()
+
+
+
 X, y = load_digits(return_X_y=True)
 X = np.asarray(X, dtype=np.float32)
 
 ovr = DistOneVsRestClassifier(
-    LogisticRegression(epochs=30, random_state=0), norm="l1", sc=None)
+    LogisticRegression(epochs=30, random_state=0), norm="l1", sc=_sc())
 ovr.fit(X, y)
 print("OvR F1w:", round(f1_score(y, ovr.predict(X), average="weighted"), 4))
 
 ovo = DistOneVsOneClassifier(
-    LogisticRegression(epochs=30, random_state=0), sc=None)
+    LogisticRegression(epochs=30, random_state=0), sc=_sc())
 ovo.fit(X, y)
 print("OvO F1w:", round(f1_score(y, ovo.predict(X), average="weighted"), 4))

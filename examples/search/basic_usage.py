@@ -23,7 +23,9 @@ data = load_breast_cancer()
 X = np.asarray(data.data, dtype=np.float32)
 y = data.target
 
-sc = None  # or Cluster() on a GPU node / under torchrun
+import torch
+
+sc = Cluster() if torch.cuda.is_available() else None
 gs = DistGridSearchCV(
     LogisticRegression(epochs=30, random_state=0),
     {"C": [0.001, 0.01, 0.1, 1.0, 10.0, 100.0]},

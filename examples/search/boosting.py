@@ -12,6 +12,19 @@ from sklearn.datasets import load_breast_cancer
 from skdist_amd.distribute.search import DistRandomizedSearchCV
 from skdist_amd.models import HistGradientBoostingClassifier
 
+
+def _sc():
+    """Cluster() on a GPU node, None for the local CPU path."""
+    import torch
+
+    if not torch.cuda.is_available():
+        return None
+    from skdist_amd import Cluster
+
+    return Cluster()
+
+
+
 X, y = load_breast_cancer(return_X_y=True)
 
 search = DistRandomizedSearchCV(
@@ -23,7 +36,7 @@ search = DistRandomizedSearchCV(
         "learning_rate": [0.05, 0.1, 0.3],
         "subsample": [0.7, 1.0],
     },
-    n_iter=4, cv=3, scoring="roc_auc", random_state=0, sc=None,
+    n_iter=4, cv=3, scoring="roc_auc", random_state=0, sc=_sc(),
 )
 search.fit(X.astype(np.float32), y)
 print("best params:", search.best_params_)

@@ -11,6 +11,19 @@ from sklearn.datasets import load_breast_cancer
 from skdist_amd.distribute.search import DistMultiModelSearch
 from skdist_amd.models import LinearSVC, LogisticRegression
 
+
+def _sc():
+    """Cluster() on a GPU node, None for the local CPU path."""
+    import torch
+
+    if not torch.cuda.is_available():
+        return None
+    from skdist_amd import Cluster
+
+    return Cluster()
+
+
+
 X, y = load_breast_cancer(return_X_y=True)
 X = np.asarray(X, dtype=np.float32)
 
@@ -21,7 +34,7 @@ search = DistMultiModelSearch(
         ("svc", LinearSVC(epochs=20, random_state=0),
          {"C": [0.01, 0.1, 1.0]}),
     ],
-    n=3, cv=3, sc=None, random_state=0)
+    n=3, cv=3, sc=_sc(), random_state=0)
 search.fit(X, y)
 print(search.results_table_ if hasattr(search, "results_table_")
       else search.cv_results_)

@@ -12,6 +12,19 @@ from sklearn.preprocessing import StandardScaler
 from skdist_amd.distribute.search import DistGridSearchCV
 from skdist_amd.models import LogisticRegression
 
+
+def _sc():
+    """Cluster() on a GPU node, None for the local CPU path."""
+    import torch
+
+    if not torch.cuda.is_available():
+        return None
+    from skdist_amd import Cluster
+
+    return Cluster()
+
+
+
 X, y = load_breast_cancer(return_X_y=True)
 X = np.asarray(X, dtype=np.float32)
 
@@ -19,7 +32,7 @@ X = np.asarray(X, dtype=np.float32)
 pipe_in_search = DistGridSearchCV(
     Pipeline([("scale", StandardScaler()),
               ("clf", LogisticRegression(epochs=20, random_state=0))]),
-    {"clf__C": [0.1, 1.0, 10.0]}, cv=5, scoring="roc_auc", sc=None)
+    {"clf__C": [0.1, 1.0, 10.0]}, cv=5, scoring="roc_auc", sc=_sc())
 pipe_in_search.fit(X, y)
 print("pipeline-in-search best CV:", round(pipe_in_search.best_score_, 5))
 
@@ -28,7 +41,7 @@ search_in_pipe = Pipeline([
     ("scale", StandardScaler()),
     ("search", DistGridSearchCV(
         LogisticRegression(epochs=20, random_state=0),
-        {"C": [0.1, 1.0, 10.0]}, cv=5, scoring="roc_auc", sc=None)),
+        {"C": [0.1, 1.0, 10.0]}, cv=5, scoring="roc_auc", sc=_sc())),
 ])
 search_in_pipe.fit(X, y)
 print("search-in-pipeline best CV:",
