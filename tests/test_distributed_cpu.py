@@ -328,6 +328,26 @@ def _sparse_text_data():
     return X, y
 
 
+def _scenario_weighted(rank):
+    """Sliced fit-param sample_weight through the sharded generic path
+    (round-2 audit fix), world 2."""
+    from sklearn.linear_model import LogisticRegression as SkLR
+
+    from skdist_amd import Cluster
+    from skdist_amd.distribute.search import DistGridSearchCV
+
+    sc = Cluster()
+    rng = np.random.default_rng(0)
+    X = rng.standard_normal((300, 5))
+    y = (X[:, 0] > 0).astype(int)
+    w = rng.random(300)
+    gs = DistGridSearchCV(SkLR(max_iter=100),
+                          {"C": [0.1, 1.0, 10.0]}, cv=3, sc=sc)
+    gs.fit(X, y, sample_weight=w)
+    return {"scores": list(gs.cv_results_["mean_test_score"]),
+            "best": gs.best_score_}
+
+
 def _scenario_sparse(rank):
     """Rank-0-only sparse CSR data through the broadcast + sparse-native
     batched solve (round-2 path; never multi-process before this)."""
@@ -354,6 +374,7 @@ def _scenario_sparse(rank):
 
 
 _SCENARIOS = {
+    "weighted": _scenario_weighted,
     "sparse": _scenario_sparse,
     "ridge": _scenario_ridge,
     "ovo": _scenario_ovo,
@@ -421,6 +442,24 @@ def test_spmd_batched_gloo():
         outs[0]["scores"], list(gs.cv_results_["mean_test_score"]),
         rtol=0, atol=1e-12,
     )
+
+
+@pytest.mark.timeout(900)
+def test_spmd_weighted_fitparams_gloo():
+    outs = _run_spmd("weighted")
+    assert np.allclose(outs[0]["scores"], outs[1]["scores"])
+    # matches sklearn exactly (same assertion as the world-1 test)
+    from sklearn.linear_model import LogisticRegression as SkLR
+    from sklearn.model_selection import GridSearchCV
+
+    rng = np.random.default_rng(0)
+    X = rng.standard_normal((300, 5))
+    y = (X[:, 0] > 0).astype(int)
+    w = rng.random(300)
+    sk = GridSearchCV(SkLR(max_iter=100), {"C": [0.1, 1.0, 10.0]},
+                      cv=3).fit(X, y, sample_weight=w)
+    np.testing.assert_allclose(
+        outs[0]["scores"], sk.cv_results_["mean_test_score"], atol=1e-13)
 
 
 @pytest.mark.timeout(900)
